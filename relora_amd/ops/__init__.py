@@ -1,0 +1,15 @@
+from relora_amd.ops.functional import (  # noqa: F401
+    build_rope_cache,
+    flash_attention,
+    fused_cross_entropy,
+    layernorm,
+    lora_linear,
+    rmsnorm,
+    rmsnorm_torch,
+    rope,
+    rope_torch,
+    rotate_half,
+    swiglu,
+    swiglu_torch,
+)
+from relora_amd.ops import hip  # noqa: F401
