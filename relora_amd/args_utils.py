@@ -13,13 +13,15 @@ import yaml
 from relora_amd.utils.logging import logger
 
 
-def check_args_torchrun_main(args):
+def check_args_torchrun_main(args, n_cli_args=None):
     if args.training_config is not None:
         logger.info(
             f"Yaml config provided for the run. The file {args.training_config} is used to provide all the parameters."
         )
-        if len(sys.argv) > 3:
-            logger.error(f"argv length is {len(sys.argv)}")
+        if n_cli_args is None:
+            n_cli_args = len(sys.argv) - 1
+        if n_cli_args > 2:
+            logger.error(f"got {n_cli_args} command line arguments")
             raise RuntimeError(
                 "You provided both a yaml config and command line arguments. "
                 "Please use only one of the two options."
@@ -50,6 +52,12 @@ def check_args_torchrun_main(args):
 
     if args.tags is not None and isinstance(args.tags, str):
         args.tags = args.tags.split(",")
+
+    # NOTE: the reference performs this check after zeroing args.relora for
+    # non-peft runs, making it unreachable; we apply the documented intent.
+    if args.relora and not args.use_peft:
+        logger.warning("--relora assumes --use_peft. Setting --use_peft=True")
+        args.use_peft = True
 
     if not args.use_peft:
         # just for more clear hparam logging
@@ -82,10 +90,6 @@ def check_args_torchrun_main(args):
             "reset_optimizer_on_relora, optimizer_random_pruning and "
             "optimizer_magnitude_pruning are mutually exclusive"
         )
-
-    if args.relora and not args.use_peft:
-        logger.warning("--relora assumes --use_peft. Setting --use_peft=True")
-        args.use_peft = True
 
     assert 0 <= args.optimizer_random_pruning < 1, "--optimizer_random_pruning must be between 0 and 1"
     assert 0 <= args.optimizer_magnitude_pruning < 1, "--optimizer_magnitude_pruning must be between 0 and 1"

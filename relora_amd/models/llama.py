@@ -114,9 +114,10 @@ class LlamaAttention(nn.Module):
             v = torch.cat([past_key_value[1], v], dim=2)
         present = (k, v) if use_cache else None
 
-        # parity with the reference: attention is ALWAYS causal; padding masks
-        # are ignored (reference modeling_llama.py:221-224)
-        attn = ops.flash_attention(q, k, v, causal=True)
+        # parity with the reference: attention is ALWAYS causal in training;
+        # padding masks are ignored (reference modeling_llama.py:221-224).
+        # q_len==1 (cached decode) must not be top-left-aligned causal.
+        attn = ops.flash_attention(q, k, v, causal=q_len > 1)
         attn = attn.transpose(1, 2).reshape(bsz, q_len, self.hidden_size)
         return self.o_proj(attn), present
 

@@ -30,8 +30,17 @@ class GPTNeoXRotary(nn.Module):
         self.dim = dim
         self.base = base
         self.max_position_embeddings = max_position_embeddings
-        self.scaling_type = scaling["type"] if scaling else None
-        self.scaling_factor = scaling["factor"] if scaling else 1.0
+        # transformers' PretrainedConfig normalizes rope_scaling to
+        # {'rope_type': 'default', ...}; treat 'default' as no scaling
+        stype = None
+        sfactor = 1.0
+        if scaling:
+            stype = scaling.get("type", scaling.get("rope_type"))
+            if stype == "default":
+                stype = None
+            sfactor = scaling.get("factor", 1.0)
+        self.scaling_type = stype
+        self.scaling_factor = sfactor
         self.max_seq_len_cached = 0
         self._build(max_position_embeddings, None)
 

@@ -44,6 +44,7 @@ class ZeroRedundancyAdamW:
             self.owner[i] = r
             loads[r] += self.all_params[i].numel()
 
+        self._param_index = {id(p): i for i, p in enumerate(self.all_params)}
         self.shard_params = [p for p, o in zip(self.all_params, self.owner) if o == self.rank]
         self.optim = AdamW(self.shard_params if self.shard_params else
                            [torch.nn.Parameter(torch.zeros(1))], **adamw_kwargs)
@@ -125,7 +126,7 @@ class ZeroRedundancyAdamW:
         local = {}
         for p in self.shard_params:
             st = self.optim.state.get(p, {})
-            idx = self.all_params.index(p)
+            idx = self._param_index[id(p)]
             local[idx] = {k: (v.cpu() if torch.is_tensor(v) else v) for k, v in st.items()}
         if not _dist_active():
             self._consolidated = [local]
@@ -143,7 +144,7 @@ class ZeroRedundancyAdamW:
                 merged.update(d)
         else:
             for p in self.shard_params:
-                idx = self.all_params.index(p)
+                idx = self._param_index[id(p)]
                 st = self.optim.state.get(p, {})
                 merged[idx] = {k: (v.cpu() if torch.is_tensor(v) else v) for k, v in st.items()}
         group = dict(self.optim.param_groups[0])

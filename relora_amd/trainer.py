@@ -135,8 +135,9 @@ def parse_args(args=None):
 
     parser.add_argument("--seed", type=int, default=0)
 
+    n_cli = len(args) if args is not None else None
     args = parser.parse_args(args)
-    args = args_utils.check_args_torchrun_main(args)
+    args = args_utils.check_args_torchrun_main(args, n_cli_args=n_cli)
     return args
 
 
@@ -263,12 +264,10 @@ def main(args):
 
     logger.info(f"Global rank {global_rank}, local rank {local_rank}, device: {device}")
 
-    if not dist.is_initialized() and "MASTER_ADDR" in os.environ:
-        dist.init_process_group(backend=backend, rank=global_rank, world_size=world_size)  # C1
-    elif not dist.is_initialized():
+    if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
-        dist.init_process_group(backend=backend, rank=global_rank, world_size=world_size)
+        dist.init_process_group(backend=backend, rank=global_rank, world_size=world_size)  # C1
 
     if args.total_batch_size is not None and args.gradient_accumulation is None:
         assert args.total_batch_size % world_size == 0, "total_batch_size must be divisible by world_size"

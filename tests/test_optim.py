@@ -1,0 +1,85 @@
+"""Fused AdamW / clip_grad_norm oracle tests vs torch.optim equivalents."""
+
+import pytest
+import torch
+
+from relora_amd.ops.optim import AdamW, clip_grad_norm_
+
+
+def _make(n=3, dtype=torch.float32, seed=0):
+    torch.manual_seed(seed)
+    params = [torch.nn.Parameter(torch.randn(17 * (i + 1), dtype=dtype)) for i in range(n)]
+    grads = [torch.randn_like(p) for p in params]
+    return params, grads
+
+
+def test_adamw_matches_torch():
+    params, grads = _make()
+    ref_params = [torch.nn.Parameter(p.detach().clone()) for p in params]
+
+    opt = AdamW(params, lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+    ref = torch.optim.AdamW(ref_params, lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.01)
+
+    for step in range(5):
+        for p, rp, g in zip(params, ref_params, grads):
+            p.grad = g * (step + 1)
+            rp.grad = g * (step + 1)
+        opt.step()
+        ref.step()
+
+    for p, rp in zip(params, ref_params):
+        assert torch.allclose(p, rp, atol=1e-6), (p - rp).abs().max()
+
+
+def test_adamw_state_layout():
+    params, grads = _make(1)
+    opt = AdamW(params, lr=1e-3)
+    params[0].grad = grads[0]
+    opt.step()
+    state = opt.state[params[0]]
+    assert "exp_avg" in state and "exp_avg_sq" in state
+    assert state["exp_avg"].shape == params[0].shape
+
+
+def test_adamw_zero_wd():
+    params, grads = _make(1)
+    ref_params = [torch.nn.Parameter(p.detach().clone()) for p in params]
+    opt = AdamW(params, lr=1e-2, weight_decay=0.0)
+    ref = torch.optim.AdamW(ref_params, lr=1e-2, weight_decay=0.0)
+    for p, rp, g in zip(params, ref_params, grads):
+        p.grad = g
+        rp.grad = g
+    opt.step()
+    ref.step()
+    assert torch.allclose(params[0], ref_params[0], atol=1e-7)
+
+
+def test_clip_grad_norm_matches_torch():
+    params, grads = _make()
+    for p, g in zip(params, grads):
+        p.grad = g.clone() * 10
+    ref_params = [torch.nn.Parameter(p.detach().clone()) for p in params]
+    for rp, g in zip(ref_params, grads):
+        rp.grad = g.clone() * 10
+
+    norm = clip_grad_norm_(params, 1.0)
+    ref_norm = torch.nn.utils.clip_grad_norm_(ref_params, 1.0)
+    assert torch.allclose(norm, ref_norm, atol=1e-5)
+    for p, rp in zip(params, ref_params):
+        assert torch.allclose(p.grad, rp.grad, atol=1e-6)
+
+
+def test_clip_grad_norm_nonfinite_raises():
+    params, grads = _make(1)
+    params[0].grad = grads[0]
+    params[0].grad[0] = float("nan")
+    with pytest.raises(RuntimeError):
+        clip_grad_norm_(params, 1.0, error_if_nonfinite=True)
+
+
+def test_clip_no_scale_when_below():
+    params, grads = _make(1)
+    params[0].grad = grads[0] * 1e-6
+    before = params[0].grad.clone()
+    clip_grad_norm_(params, 1.0)
+    assert torch.equal(params[0].grad, before)

@@ -1,0 +1,182 @@
+"""ReLoRA adapter semantics (reference relora.py parity): zero-init
+equivalence at wrap, merge idempotence, reinit distribution, trainable-param
+accounting, save/from_pretrained roundtrip, optimizer reset pruning."""
+
+import copy
+import math
+
+import pytest
+import torch
+
+from relora_amd.models.llama import LlamaForCausalLM
+from relora_amd.relora import ReLoRaLinear, ReLoRaModel, merge_and_reinit_functional
+from relora_amd.training_utils import magnitude_pruning_, optimizer_reset, random_pruning_
+
+
+def make_wrapped(config, r=8, dropout=0.0, **kw):
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(config)
+    return ReLoRaModel(
+        model, r=r, lora_alpha=32, lora_dropout=dropout,
+        target_modules=["attn", "attention", "mlp"], keep_original_weights=True, **kw
+    )
+
+
+def test_zero_init_equivalence(tiny_llama_config):
+    torch.manual_seed(0)
+    ref = LlamaForCausalLM(tiny_llama_config)
+    model = copy.deepcopy(ref)
+    wrapped = ReLoRaModel(
+        model, r=8, lora_alpha=32, lora_dropout=0.0,
+        target_modules=["attn", "attention", "mlp"], keep_original_weights=True,
+    )
+    x = torch.randint(0, tiny_llama_config.vocab_size, (2, 32))
+    with torch.no_grad():
+        l0 = ref(input_ids=x, labels=x).loss
+        l1 = wrapped(input_ids=x, labels=x).loss
+    assert torch.equal(l0, l1)
+
+
+def test_merge_idempotence(tiny_llama_config):
+    wrapped = make_wrapped(tiny_llama_config)
+    with torch.no_grad():
+        for n, p in wrapped.named_parameters():
+            if "lora_" in n:
+                p.add_(torch.randn_like(p) * 0.02)
+    x = torch.randint(0, tiny_llama_config.vocab_size, (2, 32))
+    wrapped.eval()
+    with torch.no_grad():
+        pre = wrapped(input_ids=x, labels=x).loss
+    wrapped.merge_and_reinit()
+    with torch.no_grad():
+        post = wrapped(input_ids=x, labels=x).loss
+    assert torch.allclose(pre, post, atol=1e-5), (pre, post)
+    # after merge, B is zero and A is re-initialized (non-zero)
+    for m in wrapped.modules():
+        if isinstance(m, ReLoRaLinear):
+            assert torch.all(m.lora_B.weight == 0)
+            assert m.lora_A.weight.abs().sum() > 0
+
+
+def test_functional_merge_matches_method(tiny_llama_config):
+    w1 = make_wrapped(tiny_llama_config)
+    with torch.no_grad():
+        for n, p in w1.named_parameters():
+            if "lora_" in n:
+                p.add_(torch.randn_like(p) * 0.02)
+    w2 = copy.deepcopy(w1)
+    w1.merge_and_reinit()
+    for m in w2.modules():
+        merge_and_reinit_functional(m)
+    for (n1, p1), (n2, p2) in zip(w1.named_parameters(), w2.named_parameters()):
+        if "lora_" not in n1:
+            assert torch.equal(p1, p2), n1
+
+
+def test_trainable_params_accounting(tiny_llama_config):
+    wrapped = make_wrapped(tiny_llama_config, r=8)
+    H, inter = tiny_llama_config.hidden_size, tiny_llama_config.intermediate_size
+    L = tiny_llama_config.num_hidden_layers
+    # per layer: q,k,v,o (HxH) + gate,up (inter x H) + down (H x inter)
+    expected_lora = L * (4 * 8 * (H + H) + 2 * 8 * (H + inter) + 8 * (inter + H))
+    got_lora = sum(
+        p.numel() for n, p in wrapped.named_parameters() if "lora_" in n
+    )
+    assert got_lora == expected_lora
+    # frozen: exactly the wrapped W matrices
+    frozen = [n for n, p in wrapped.named_parameters() if not p.requires_grad]
+    assert all(n.endswith(".weight") and "lora" not in n for n in frozen)
+    assert len(frozen) == L * 7
+
+
+def test_trainable_scaling(tiny_llama_config):
+    wrapped = make_wrapped(tiny_llama_config, trainable_scaling=True)
+    for m in wrapped.modules():
+        if isinstance(m, ReLoRaLinear):
+            assert isinstance(m.scaling, torch.nn.Parameter)
+            assert float(m._post_lora_scale().detach()) == pytest.approx(math.tanh(1.0))
+    wrapped.merge_and_reinit()
+    for m in wrapped.modules():
+        if isinstance(m, ReLoRaLinear):
+            assert m.scaling.item() == 0.0
+
+
+def test_save_load_roundtrip(tiny_llama_config, tmp_path):
+    wrapped = make_wrapped(tiny_llama_config)
+    with torch.no_grad():
+        for n, p in wrapped.named_parameters():
+            if "lora_" in n:
+                p.add_(torch.randn_like(p) * 0.02)
+    wrapped.save_pretrained(tmp_path / "ckpt")
+    assert (tmp_path / "ckpt" / "pytorch_model.bin").exists()
+    assert (tmp_path / "ckpt" / "relora_config.json").exists()
+    assert (tmp_path / "ckpt" / "config.json").exists()
+
+    reloaded = ReLoRaModel.from_pretrained(str(tmp_path / "ckpt"))
+    x = torch.randint(0, tiny_llama_config.vocab_size, (2, 16))
+    wrapped.eval()
+    reloaded.eval()
+    with torch.no_grad():
+        a = wrapped(input_ids=x, labels=x).loss
+        b = reloaded(input_ids=x, labels=x).loss
+    assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_r_zero_raises(tiny_llama_config):
+    model = LlamaForCausalLM(tiny_llama_config)
+    with pytest.raises(ValueError):
+        ReLoRaModel(model, r=0, target_modules=["attn"])
+
+
+def test_quantize_raises(tiny_llama_config):
+    model = LlamaForCausalLM(tiny_llama_config)
+    with pytest.raises(NotImplementedError):
+        ReLoRaModel(model, r=8, target_modules=["attn"], quantize="4bit")
+
+
+# ---------------------------------------------------------------------------
+# optimizer reset / pruning
+# ---------------------------------------------------------------------------
+
+
+def test_random_pruning_ratio():
+    torch.manual_seed(0)
+    t = torch.randn(100_000)
+    random_pruning_(t, 0.7)
+    frac = (t == 0).float().mean().item()
+    assert 0.68 < frac < 0.72
+
+
+def test_magnitude_pruning_keeps_largest():
+    torch.manual_seed(0)
+    t = torch.randn(10_000)
+    orig = t.clone()
+    magnitude_pruning_(t, 0.8)
+    frac = (t == 0).float().mean().item()
+    assert 0.79 < frac < 0.81
+    kept = t != 0
+    # every kept value is larger in magnitude than every dropped original
+    assert orig[kept].abs().min() >= orig[~kept].abs().max() - 1e-6
+
+
+def test_optimizer_reset_modes(tiny_llama_config):
+    wrapped = make_wrapped(tiny_llama_config)
+    params = [p for n, p in wrapped.named_parameters() if p.requires_grad and "lora_" in n]
+    opt = torch.optim.AdamW(params, lr=1e-3)
+    x = torch.randint(0, tiny_llama_config.vocab_size, (2, 16))
+    wrapped(input_ids=x, labels=x).loss.backward()
+    opt.step()
+
+    zeroed = optimizer_reset(
+        opt, reset_params=params, optimizer_state_keys=["exp_avg", "exp_avg_sq"],
+        reset_optimizer_on_relora=False, optimizer_random_pruning=0.0,
+        optimizer_magnitude_pruning=0.9,
+    )
+    assert zeroed > 85
+
+    with pytest.raises(ValueError):
+        optimizer_reset(
+            opt, reset_params=params, optimizer_state_keys=["exp_avg"],
+            reset_optimizer_on_relora=True, optimizer_random_pruning=0.5,
+            optimizer_magnitude_pruning=0.0,
+        )

@@ -1,0 +1,158 @@
+"""End-to-end trainer tests on CPU synthetic data: loop runs, resets fire,
+checkpoints have the reference layout, autoresume restores counters, args
+validation matches the reference contract."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from relora_amd.trainer import main, parse_args
+
+
+def run_args(tmp_path, extra=None, steps=6):
+    base = [
+        "--model_config", "configs/llama_9m.json",
+        "--synthetic_data", "true",
+        "--use_peft", "true",
+        "--relora", "3", "--cycle_length", "3",
+        "--restart_warmup_steps", "1",
+        "--scheduler", "cosine_restarts",
+        "--warmup_steps", "2",
+        "--num_training_steps", str(steps),
+        "--batch_size", "2", "--total_batch_size", "4",
+        "--max_length", "32",
+        "--lr", "1e-3", "--dtype", "float32",
+        "--eval_every", "100", "--save_every", "100",
+        "--workers", "0",
+        "--save_dir", str(tmp_path / "run"),
+    ]
+    return parse_args(base + (extra or []))
+
+
+@pytest.fixture(autouse=True)
+def _single_proc_env(monkeypatch):
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT"):
+        monkeypatch.delenv(k, raising=False)
+    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+    yield
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def test_trainer_end_to_end(tmp_path):
+    args = run_args(tmp_path)
+    main(args)
+    ckpt = tmp_path / "run" / "model_6"
+    assert ckpt.exists()
+    for f in ("pytorch_model.bin", "config.json", "relora_config.json",
+              "optimizer.pt", "training_state.json"):
+        assert (ckpt / f).exists(), f
+    state = json.load(open(ckpt / "training_state.json"))
+    assert state["update_step"] == 6
+    assert state["n_lora_restarts"] >= 1
+    assert state["n_optimizer_resets"] >= 1
+    assert (tmp_path / "run" / "training_config.yaml").exists()
+    opt_ckpt = torch.load(ckpt / "optimizer.pt", map_location="cpu", weights_only=False)
+    assert "optimizer" in opt_ckpt and "scheduler" in opt_ckpt
+    assert opt_ckpt["update_step"] == 6
+
+
+def test_trainer_autoresume(tmp_path):
+    main(run_args(tmp_path, steps=6))
+    # resume to 9 steps (num_training_steps must stay divisible by cycle_length)
+    args = run_args(tmp_path, extra=["--autoresume", "true"], steps=9)
+    main(args)
+    state = json.load(open(tmp_path / "run" / "model_9" / "training_state.json"))
+    assert state["update_step"] == 9
+    assert state["tokens_seen"] > 0
+
+
+def test_trainer_zero_optimizer(tmp_path):
+    args = run_args(tmp_path, extra=["--optimizer", "adam_zero"])
+    main(args)
+    assert (tmp_path / "run" / "model_6" / "optimizer.pt").exists()
+
+
+def test_trainer_full_rank(tmp_path):
+    base = [
+        "--model_config", "configs/llama_9m.json",
+        "--synthetic_data", "true",
+        "--num_training_steps", "3",
+        "--batch_size", "2", "--total_batch_size", "2",
+        "--max_length", "32", "--lr", "1e-3", "--dtype", "float32",
+        "--eval_every", "100", "--save_every", "100", "--workers", "0",
+        "--save_dir", str(tmp_path / "runf"),
+    ]
+    args = parse_args(base)
+    assert args.use_peft is False
+    main(args)
+    assert (tmp_path / "runf" / "model_3" / "pytorch_model.bin").exists()
+
+
+# ---------------------------------------------------------------------------
+# args contract
+# ---------------------------------------------------------------------------
+
+
+def test_args_batch_algebra():
+    args = parse_args([
+        "--synthetic_data", "true", "--batch_size", "4", "--gradient_accumulation", "3",
+    ])
+    assert args.total_batch_size == 12
+
+
+def test_args_requires_batch_size():
+    with pytest.raises(ValueError):
+        parse_args(["--synthetic_data", "true"])
+
+
+def test_args_data_source_exclusive():
+    with pytest.raises(ValueError):
+        parse_args(["--batch_size", "2"])
+    with pytest.raises(ValueError):
+        parse_args(["--synthetic_data", "true", "--dataset_path", "/tmp/x",
+                    "--batch_size", "2"])
+
+
+def test_args_fp16_rejected():
+    with pytest.raises(NotImplementedError):
+        parse_args(["--synthetic_data", "true", "--batch_size", "2", "--dtype", "float16"])
+
+
+def test_args_relora_implies_peft():
+    args = parse_args(["--synthetic_data", "true", "--batch_size", "2", "--relora", "10"])
+    assert args.use_peft is True
+
+
+def test_args_reset_modes_exclusive():
+    with pytest.raises(ValueError):
+        parse_args([
+            "--synthetic_data", "true", "--batch_size", "2",
+            "--reset_optimizer_on_relora", "true",
+            "--optimizer_magnitude_pruning", "0.9",
+        ])
+
+
+def test_args_skip_batches_parsing():
+    args = parse_args([
+        "--synthetic_data", "true", "--batch_size", "2", "--skip_batches", "3,5,9",
+    ])
+    assert args.skip_batches == {3, 5, 9}
+
+
+def test_args_yaml_override(tmp_path):
+    import yaml
+
+    cfg = {
+        "synthetic_data": True, "batch_size": 2, "total_batch_size": 4,
+        "lr": "5e-4", "num_training_steps": 10,
+    }
+    path = tmp_path / "cfg.yaml"
+    path.write_text(yaml.dump(cfg))
+    args = parse_args(["--training_config", str(path)])
+    assert args.batch_size == 2
+    assert args.lr == 5e-4
+    assert isinstance(args.lr, float)
