@@ -1,0 +1,54 @@
+// Python bindings for the relora_amd gfx950 HIP kernels.
+
+#include <torch/extension.h>
+
+// norms.hip
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor invrms, torch::Tensor dy);
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor mean, torch::Tensor invstd,
+                                         torch::Tensor dy);
+// rope.hip
+std::vector<torch::Tensor> rope_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor cos_t, torch::Tensor sin_t,
+                                    bool inverse);
+// swiglu.hip
+torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u);
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u, torch::Tensor dy);
+// ce.hip
+std::vector<torch::Tensor> ce_row_stats(torch::Tensor logits, torch::Tensor labels,
+                                        long ignore_index);
+void ce_grad_(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+              double gscale, long ignore_index);
+// adamw.hip
+void fused_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+                 double lr, double beta1, double beta2, double eps, double wd, long step);
+torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> grads);
+void multi_tensor_scale_(std::vector<torch::Tensor> grads, double scale);
+// attention.hip
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    double scale);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse, torch::Tensor dout,
+                                    double scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (gfx950)");
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (gfx950)");
+  m.def("rope_fwd", &rope_fwd, "RoPE apply fwd/inverse (gfx950)");
+  m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (gfx950)");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (gfx950)");
+  m.def("ce_row_stats", &ce_row_stats, "CE row lse/target (gfx950)");
+  m.def("ce_grad_", &ce_grad_, "CE in-place softmax-onehot grad (gfx950)");
+  m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (gfx950)");
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "multi-tensor L2 norm (gfx950)");
+  m.def("multi_tensor_scale_", &multi_tensor_scale_, "multi-tensor scale (gfx950)");
+  m.def("attn_fwd", &attn_fwd, "causal flash attention forward (gfx950 MFMA)");
+  m.def("attn_bwd", &attn_bwd, "causal flash attention backward (gfx950 MFMA)");
+}

@@ -1,0 +1,351 @@
+#include "hip/hip_runtime.h"
+// RMSNorm (K5) and LayerNorm (K6) for gfx950.
+// Reference numerics: fp32 variance, product rounded at bf16
+// (reference modeling_llama.py:74-91).
+//
+// fwd: one block per row, vectorized 8-wide loads, single pass.
+// bwd: dx one block per row; dw/db one thread per column striding rows
+// (coalesced across threads; the re-read of x/dy is HBM-bound and overlaps).
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// RMSNorm forward: y = w * (x * rsqrt(mean(x^2)+eps)); saves invrms per row.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   T* __restrict__ y, float* __restrict__ invrms,
+                                   int H, float eps) {
+  __shared__ float scratch[16];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)H;
+  T* yr = y + row * (long)H;
+
+  float ss = 0.f;
+  const int vec_end = (H / 8) * 8;
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> v = load8(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(v.v[j]);
+      ss += f * f;
+    }
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x) {
+    float f = to_f32(xr[i]);
+    ss += f * f;
+  }
+  ss = block_reduce_sum(ss, scratch);
+  const float ir = rsqrtf(ss / H + eps);
+  if (threadIdx.x == 0) invrms[row] = ir;
+
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> v = load8(xr + i);
+    Vec8<T> wv = load8(w + i);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // match torch: (x * invrms) rounded to T, then * w rounded to T
+      T h = from_f32<T>(to_f32(v.v[j]) * ir);
+      o.v[j] = from_f32<T>(to_f32(h) * to_f32(wv.v[j]));
+    }
+    store8(yr + i, o);
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x) {
+    T h = from_f32<T>(to_f32(xr[i]) * ir);
+    yr[i] = from_f32<T>(to_f32(h) * to_f32(w[i]));
+  }
+}
+
+// dx = invrms * (g - xhat * mean(g * xhat)), g = dy*w, xhat = x*invrms
+template <typename T>
+__global__ void rmsnorm_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                      const float* __restrict__ invrms,
+                                      const T* __restrict__ dy, T* __restrict__ dx,
+                                      int H) {
+  __shared__ float scratch[16];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)H;
+  const T* dyr = dy + row * (long)H;
+  T* dxr = dx + row * (long)H;
+  const float ir = invrms[row];
+
+  float dot = 0.f;
+  const int vec_end = (H / 8) * 8;
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> xv = load8(xr + i);
+    Vec8<T> dv = load8(dyr + i);
+    Vec8<T> wv = load8(w + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dot += to_f32(dv.v[j]) * to_f32(wv.v[j]) * to_f32(xv.v[j]) * ir;
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x)
+    dot += to_f32(dyr[i]) * to_f32(w[i]) * to_f32(xr[i]) * ir;
+  dot = block_reduce_sum(dot, scratch) / H;  // mean(g * xhat)
+
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> xv = load8(xr + i);
+    Vec8<T> dv = load8(dyr + i);
+    Vec8<T> wv = load8(w + i);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = to_f32(dv.v[j]) * to_f32(wv.v[j]);
+      float xh = to_f32(xv.v[j]) * ir;
+      o.v[j] = from_f32<T>(ir * (g - xh * dot));
+    }
+    store8(dxr + i, o);
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x) {
+    float g = to_f32(dyr[i]) * to_f32(w[i]);
+    float xh = to_f32(xr[i]) * ir;
+    dxr[i] = from_f32<T>(ir * (g - xh * dot));
+  }
+}
+
+// dw[col] = sum_rows dy[r][col] * x[r][col] * invrms[r]  (fp32 out)
+template <typename T>
+__global__ void rmsnorm_bwd_dw_kernel(const T* __restrict__ x,
+                                      const float* __restrict__ invrms,
+                                      const T* __restrict__ dy,
+                                      float* __restrict__ dw, long M, int H) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  float acc = 0.f;
+  for (long r = 0; r < M; ++r) {
+    acc += to_f32(dy[r * H + col]) * to_f32(x[r * H + col]) * invrms[r];
+  }
+  dw[col] = acc;
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void layernorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                     const T* __restrict__ b, T* __restrict__ y,
+                                     float* __restrict__ mean_out,
+                                     float* __restrict__ invstd_out, int H, float eps) {
+  __shared__ float scratch[16];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)H;
+  T* yr = y + row * (long)H;
+
+  float s = 0.f, ss = 0.f;
+  const int vec_end = (H / 8) * 8;
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> v = load8(xr + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = to_f32(v.v[j]);
+      s += f;
+      ss += f * f;
+    }
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x) {
+    float f = to_f32(xr[i]);
+    s += f;
+    ss += f * f;
+  }
+  s = block_reduce_sum(s, scratch);
+  ss = block_reduce_sum(ss, scratch);
+  const float mu = s / H;
+  const float var = fmaxf(ss / H - mu * mu, 0.f);
+  const float istd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean_out[row] = mu;
+    invstd_out[row] = istd;
+  }
+
+  for (int i = threadIdx.x * 8; i < vec_end; i += blockDim.x * 8) {
+    Vec8<T> v = load8(xr + i);
+    Vec8<T> wv = load8(w + i);
+    Vec8<T> bv = load8(b + i);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xh = (to_f32(v.v[j]) - mu) * istd;
+      o.v[j] = from_f32<T>(xh * to_f32(wv.v[j]) + to_f32(bv.v[j]));
+    }
+    store8(yr + i, o);
+  }
+  for (int i = vec_end + threadIdx.x; i < H; i += blockDim.x) {
+    float xh = (to_f32(xr[i]) - mu) * istd;
+    yr[i] = from_f32<T>(xh * to_f32(w[i]) + to_f32(b[i]));
+  }
+}
+
+// dx = istd * (g - mean(g) - xhat * mean(g * xhat)), g = dy*w
+template <typename T>
+__global__ void layernorm_bwd_dx_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                        const float* __restrict__ mean,
+                                        const float* __restrict__ invstd,
+                                        const T* __restrict__ dy, T* __restrict__ dx,
+                                        int H) {
+  __shared__ float scratch[16];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)H;
+  const T* dyr = dy + row * (long)H;
+  T* dxr = dx + row * (long)H;
+  const float mu = mean[row];
+  const float istd = invstd[row];
+
+  float sg = 0.f, sgx = 0.f;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    float g = to_f32(dyr[i]) * to_f32(w[i]);
+    float xh = (to_f32(xr[i]) - mu) * istd;
+    sg += g;
+    sgx += g * xh;
+  }
+  sg = block_reduce_sum(sg, scratch) / H;
+  sgx = block_reduce_sum(sgx, scratch) / H;
+
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    float g = to_f32(dyr[i]) * to_f32(w[i]);
+    float xh = (to_f32(xr[i]) - mu) * istd;
+    dxr[i] = from_f32<T>(istd * (g - sg - xh * sgx));
+  }
+}
+
+template <typename T>
+__global__ void layernorm_bwd_dwdb_kernel(const T* __restrict__ x,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          const T* __restrict__ dy,
+                                          float* __restrict__ dw,
+                                          float* __restrict__ db, long M, int H) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  float accw = 0.f, accb = 0.f;
+  for (long r = 0; r < M; ++r) {
+    float d = to_f32(dy[r * H + col]);
+    float xh = (to_f32(x[r * H + col]) - mean[r]) * invstd[r];
+    accw += d * xh;
+    accb += d;
+  }
+  dw[col] = accw;
+  db[col] = accb;
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static int norm_block(int H) {
+  if (H <= 512) return 64;
+  if (H <= 2048) return 256;
+  return 512;
+}
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  const long M = x.size(0);
+  const int H = x.size(1);
+  auto y = torch::empty_like(x);
+  auto invrms = torch::empty({M}, x.options().dtype(torch::kFloat32));
+  dim3 grid(M), block(norm_block(H));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(rmsnorm_fwd_kernel<__hip_bfloat16>, grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
+                       (__hip_bfloat16*)y.data_ptr(), invrms.data_ptr<float>(), H, (float)eps);
+  } else {
+    hipLaunchKernelGGL(rmsnorm_fwd_kernel<float>, grid, block, 0, stream,
+                       x.data_ptr<float>(), w.data_ptr<float>(), y.data_ptr<float>(),
+                       invrms.data_ptr<float>(), H, (float)eps);
+  }
+  HIP_CHECK_LAST();
+  return {y, invrms};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
+                                       torch::Tensor invrms, torch::Tensor dy) {
+  const long M = x.size(0);
+  const int H = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid(M), block(norm_block(H));
+  dim3 gridc((H + 255) / 256), blockc(256);
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(rmsnorm_bwd_dx_kernel<__hip_bfloat16>, grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
+                       invrms.data_ptr<float>(), (const __hip_bfloat16*)dy.data_ptr(),
+                       (__hip_bfloat16*)dx.data_ptr(), H);
+    hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel<__hip_bfloat16>, gridc, blockc, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), invrms.data_ptr<float>(),
+                       (const __hip_bfloat16*)dy.data_ptr(), dw.data_ptr<float>(), M, H);
+  } else {
+    hipLaunchKernelGGL(rmsnorm_bwd_dx_kernel<float>, grid, block, 0, stream,
+                       x.data_ptr<float>(), w.data_ptr<float>(), invrms.data_ptr<float>(),
+                       dy.data_ptr<float>(), dx.data_ptr<float>(), H);
+    hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel<float>, gridc, blockc, 0, stream,
+                       x.data_ptr<float>(), invrms.data_ptr<float>(), dy.data_ptr<float>(),
+                       dw.data_ptr<float>(), M, H);
+  }
+  HIP_CHECK_LAST();
+  return {dx, dw};
+}
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  const long M = x.size(0);
+  const int H = x.size(1);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({M}, x.options().dtype(torch::kFloat32));
+  auto invstd = torch::empty({M}, x.options().dtype(torch::kFloat32));
+  dim3 grid(M), block(norm_block(H));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(layernorm_fwd_kernel<__hip_bfloat16>, grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
+                       (const __hip_bfloat16*)b.data_ptr(), (__hip_bfloat16*)y.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(), H, (float)eps);
+  } else {
+    hipLaunchKernelGGL(layernorm_fwd_kernel<float>, grid, block, 0, stream,
+                       x.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),
+                       y.data_ptr<float>(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       H, (float)eps);
+  }
+  HIP_CHECK_LAST();
+  return {y, mean, invstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor mean, torch::Tensor invstd,
+                                         torch::Tensor dy) {
+  const long M = x.size(0);
+  const int H = x.size(1);
+  auto dx = torch::empty_like(x);
+  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto db = torch::empty({H}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid(M), block(norm_block(H));
+  dim3 gridc((H + 255) / 256), blockc(256);
+  if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(layernorm_bwd_dx_kernel<__hip_bfloat16>, grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       (const __hip_bfloat16*)dy.data_ptr(), (__hip_bfloat16*)dx.data_ptr(), H);
+    hipLaunchKernelGGL(layernorm_bwd_dwdb_kernel<__hip_bfloat16>, gridc, blockc, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), (const __hip_bfloat16*)dy.data_ptr(),
+                       dw.data_ptr<float>(), db.data_ptr<float>(), M, H);
+  } else {
+    hipLaunchKernelGGL(layernorm_bwd_dx_kernel<float>, grid, block, 0, stream,
+                       x.data_ptr<float>(), w.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), dy.data_ptr<float>(), dx.data_ptr<float>(), H);
+    hipLaunchKernelGGL(layernorm_bwd_dwdb_kernel<float>, gridc, blockc, 0, stream,
+                       x.data_ptr<float>(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       dy.data_ptr<float>(), dw.data_ptr<float>(), db.data_ptr<float>(), M, H);
+  }
+  HIP_CHECK_LAST();
+  return {dx, dw, db};
+}

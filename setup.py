@@ -1,0 +1,40 @@
+"""Build the relora_amd package and its in-tree gfx950 HIP extension.
+
+Usage:  python setup.py build_ext --inplace
+The built .so lands at relora_amd/ops/_relora_hip*.so and ships with the
+repo snapshot (it is git-ignored but not gpurun-ignored).
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import find_packages, setup  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+CSRC = os.path.join("relora_amd", "ops", "csrc")
+
+ext = CUDAExtension(
+    name="relora_amd.ops._relora_hip",
+    sources=[
+        os.path.join(CSRC, "bindings.cpp"),
+        os.path.join(CSRC, "norms.hip"),
+        os.path.join(CSRC, "rope.hip"),
+        os.path.join(CSRC, "swiglu.hip"),
+        os.path.join(CSRC, "ce.hip"),
+        os.path.join(CSRC, "adamw.hip"),
+        os.path.join(CSRC, "attention.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17"],
+    },
+)
+
+setup(
+    name="relora_amd",
+    version="0.1.0",
+    packages=find_packages(include=["relora_amd", "relora_amd.*"]),
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension},
+)
