@@ -49,9 +49,12 @@ class LlamaRotaryEmbedding(nn.Module):
         self.register_buffer("sin_cached", sin, persistent=False)
 
     def forward(self, x, seq_len):
-        if seq_len > self.max_seq_len_cached:
-            self.max_seq_len_cached = seq_len
-            cos, sin = ops.build_rope_cache(self.dim, seq_len, self.base, x.device)
+        # Rebuild if the cache grew or a model-wide .to(dtype) cast it away
+        # from fp32 (the RoPE kernel consumes fp32 tables).
+        if seq_len > self.max_seq_len_cached or self.cos_cached.dtype != torch.float32:
+            self.max_seq_len_cached = max(seq_len, self.max_seq_len_cached)
+            cos, sin = ops.build_rope_cache(
+                self.dim, self.max_seq_len_cached, self.base, x.device)
             self.register_buffer("cos_cached", cos, persistent=False)
             self.register_buffer("sin_cached", sin, persistent=False)
         return self.cos_cached.to(x.device), self.sin_cached.to(x.device)

@@ -66,7 +66,9 @@ class GPTNeoXRotary(nn.Module):
     def forward(self, x, seq_len):
         if seq_len > self.max_seq_len_cached or (
             self.scaling_type == "dynamic" and seq_len != self.max_seq_len_cached
-        ):
+        ) or self.cos_cached.dtype != torch.float32:
+            # fp32 tables are part of the RoPE kernel contract; a model-wide
+            # .to(dtype) may have cast the buffers.
             self._build(max(seq_len, self.max_position_embeddings), x.device)
         return self.cos_cached.to(x.device), self.sin_cached.to(x.device)
 

@@ -138,6 +138,10 @@ class _HipRoPE(torch.autograd.Function):
     def forward(ctx, q, k, cos, sin):
         q = q.contiguous()
         k = k.contiguous()
+        # Kernel contract: fp32 contiguous tables (a model .to(bf16) casts
+        # registered buffers; converting [S,hd] back is noise next to the GEMMs).
+        cos = cos.float().contiguous()
+        sin = sin.float().contiguous()
         qo, ko = hip.ext().rope_fwd(q, k, cos, sin, False)
         ctx.save_for_backward(cos, sin)
         return qo, ko
