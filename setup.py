@@ -11,8 +11,20 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import find_packages, setup  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
 
 CSRC = os.path.join("relora_amd", "ops", "csrc")
+DATA_CSRC = os.path.join("relora_amd", "data", "csrc")
+
+# Host-side C++ index builders (no HIP): plain pybind11 extension.
+index_helpers_ext = Extension(
+    name="relora_amd.data._index_helpers",
+    sources=[os.path.join(DATA_CSRC, "index_helpers.cpp")],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O3", "-std=c++17", "-fvisibility=hidden"],
+    language="c++",
+)
 
 ext = CUDAExtension(
     name="relora_amd.ops._relora_hip",
@@ -35,6 +47,6 @@ setup(
     name="relora_amd",
     version="0.1.0",
     packages=find_packages(include=["relora_amd", "relora_amd.*"]),
-    ext_modules=[ext],
+    ext_modules=[ext, index_helpers_ext],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -251,6 +251,36 @@ def test_clip_grad_norm_gpu():
                           what="clipped grad")
 
 
+
+def test_multi_tensor_ops_oversized_tensors():
+    """Tensors beyond the 320-chunk table (embeddings/lm_head on llama_1b,
+    ~65M elements) must route through the grid-stride single-tensor kernels
+    (this exact case aborted the first 1B bench run)."""
+    from relora_amd.ops.optim import AdamW, clip_grad_norm_
+
+    torch.manual_seed(0)
+    big = 33 * 1024 * 1024  # > 320 * 65536 = 21M
+    params = [torch.nn.Parameter(torch.randn(n, device="cuda", dtype=torch.float32))
+              for n in (big, 4096)]
+    ref_params = [torch.nn.Parameter(p.detach().clone()) for p in params]
+    for p, rp in zip(params, ref_params):
+        g = torch.randn_like(p)
+        p.grad = g
+        rp.grad = g.clone()
+
+    norm = clip_grad_norm_(params, 1.0)
+    ref_norm = torch.nn.utils.clip_grad_norm_(ref_params, 1.0)
+    assert torch.allclose(norm.float(), ref_norm, rtol=1e-3), (norm, ref_norm)
+    assert torch.allclose(params[0].grad, ref_params[0].grad, atol=1e-5)
+
+    opt = AdamW(params, lr=1e-2)
+    ref = torch.optim.AdamW(ref_params, lr=1e-2)
+    opt.step()
+    ref.step()
+    for p, rp in zip(params, ref_params):
+        assert torch.allclose(p, rp, atol=1e-5), (p - rp).abs().max()
+
+
 # ---------------------------------------------------------------------------
 # attention
 # ---------------------------------------------------------------------------
