@@ -109,18 +109,40 @@ __global__ void rmsnorm_bwd_dx_kernel(const T* __restrict__ x, const T* __restri
 }
 
 // dw[col] = sum_rows dy[r][col] * x[r][col] * invrms[r]  (fp32 out)
+// Two stages: blockIdx.y row-chunks write fp32 partials (fills the chip —
+// the single-stage column loop left 255/256 CUs idle and was 37% of a
+// training step), then a small combine kernel sums the chunk axis.
 template <typename T>
-__global__ void rmsnorm_bwd_dw_kernel(const T* __restrict__ x,
-                                      const float* __restrict__ invrms,
-                                      const T* __restrict__ dy,
-                                      float* __restrict__ dw, long M, int H) {
+__global__ void rmsnorm_bwd_dw_partial_kernel(const T* __restrict__ x,
+                                              const float* __restrict__ invrms,
+                                              const T* __restrict__ dy,
+                                              float* __restrict__ dw_part,
+                                              long M, int H, int rows_per_chunk) {
   const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= H) return;
+  const long r0 = (long)blockIdx.y * rows_per_chunk;
+  const long r1 = min(M, r0 + rows_per_chunk);
   float acc = 0.f;
-  for (long r = 0; r < M; ++r) {
+  for (long r = r0; r < r1; ++r)
     acc += to_f32(dy[r * H + col]) * to_f32(x[r * H + col]) * invrms[r];
-  }
-  dw[col] = acc;
+  dw_part[(long)blockIdx.y * H + col] = acc;
+}
+
+__global__ void col_combine_kernel(const float* __restrict__ part,
+                                   float* __restrict__ out, int H, int nchunks,
+                                   int nout) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H * nout) return;
+  float acc = 0.f;
+  for (int c = 0; c < nchunks; ++c) acc += part[(long)c * H * nout + col];
+  out[col] = acc;
+}
+
+static int reduce_chunks(long M) {
+  // enough chunks to fill 256 CUs even when H/256 is small, few enough that
+  // the combine stays trivial
+  long c = (M + 63) / 64;
+  return (int)(c < 1 ? 1 : (c > 256 ? 256 : c));
 }
 
 // ---------------------------------------------------------------------------
@@ -213,24 +235,28 @@ __global__ void layernorm_bwd_dx_kernel(const T* __restrict__ x, const T* __rest
   }
 }
 
+// Same two-stage structure as rmsnorm's dw; partials laid out [chunk][2][H]
+// (dw plane then db plane) so one col_combine_kernel call with nout=2 sums both.
 template <typename T>
-__global__ void layernorm_bwd_dwdb_kernel(const T* __restrict__ x,
-                                          const float* __restrict__ mean,
-                                          const float* __restrict__ invstd,
-                                          const T* __restrict__ dy,
-                                          float* __restrict__ dw,
-                                          float* __restrict__ db, long M, int H) {
+__global__ void layernorm_bwd_dwdb_partial_kernel(const T* __restrict__ x,
+                                                  const float* __restrict__ mean,
+                                                  const float* __restrict__ invstd,
+                                                  const T* __restrict__ dy,
+                                                  float* __restrict__ part,
+                                                  long M, int H, int rows_per_chunk) {
   const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= H) return;
+  const long r0 = (long)blockIdx.y * rows_per_chunk;
+  const long r1 = min(M, r0 + rows_per_chunk);
   float accw = 0.f, accb = 0.f;
-  for (long r = 0; r < M; ++r) {
+  for (long r = r0; r < r1; ++r) {
     float d = to_f32(dy[r * H + col]);
     float xh = (to_f32(x[r * H + col]) - mean[r]) * invstd[r];
     accw += d * xh;
     accb += d;
   }
-  dw[col] = accw;
-  db[col] = accb;
+  part[(long)blockIdx.y * 2 * H + col] = accw;
+  part[(long)blockIdx.y * 2 * H + H + col] = accb;
 }
 
 // ---------------------------------------------------------------------------
@@ -272,23 +298,29 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
   auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 grid(M), block(norm_block(H));
-  dim3 gridc((H + 255) / 256), blockc(256);
+  const int nchunks = reduce_chunks(M);
+  const int rows_per_chunk = (int)((M + nchunks - 1) / nchunks);
+  auto part = torch::empty({nchunks, H}, x.options().dtype(torch::kFloat32));
+  dim3 gridc((H + 255) / 256, nchunks), blockc(256);
   if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(rmsnorm_bwd_dx_kernel<__hip_bfloat16>, grid, block, 0, stream,
                        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
                        invrms.data_ptr<float>(), (const __hip_bfloat16*)dy.data_ptr(),
                        (__hip_bfloat16*)dx.data_ptr(), H);
-    hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel<__hip_bfloat16>, gridc, blockc, 0, stream,
+    hipLaunchKernelGGL(rmsnorm_bwd_dw_partial_kernel<__hip_bfloat16>, gridc, blockc, 0, stream,
                        (const __hip_bfloat16*)x.data_ptr(), invrms.data_ptr<float>(),
-                       (const __hip_bfloat16*)dy.data_ptr(), dw.data_ptr<float>(), M, H);
+                       (const __hip_bfloat16*)dy.data_ptr(), part.data_ptr<float>(),
+                       M, H, rows_per_chunk);
   } else {
     hipLaunchKernelGGL(rmsnorm_bwd_dx_kernel<float>, grid, block, 0, stream,
                        x.data_ptr<float>(), w.data_ptr<float>(), invrms.data_ptr<float>(),
                        dy.data_ptr<float>(), dx.data_ptr<float>(), H);
-    hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel<float>, gridc, blockc, 0, stream,
+    hipLaunchKernelGGL(rmsnorm_bwd_dw_partial_kernel<float>, gridc, blockc, 0, stream,
                        x.data_ptr<float>(), invrms.data_ptr<float>(), dy.data_ptr<float>(),
-                       dw.data_ptr<float>(), M, H);
+                       part.data_ptr<float>(), M, H, rows_per_chunk);
   }
+  hipLaunchKernelGGL(col_combine_kernel, dim3((H + 255) / 256), blockc, 0, stream,
+                     part.data_ptr<float>(), dw.data_ptr<float>(), H, nchunks, 1);
   HIP_CHECK_LAST();
   return {dx, dw};
 }
@@ -324,28 +356,32 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor x, torch::Tensor w,
   const long M = x.size(0);
   const int H = x.size(1);
   auto dx = torch::empty_like(x);
-  auto dw = torch::empty({H}, x.options().dtype(torch::kFloat32));
-  auto db = torch::empty({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 grid(M), block(norm_block(H));
-  dim3 gridc((H + 255) / 256), blockc(256);
+  const int nchunks = reduce_chunks(M);
+  const int rows_per_chunk = (int)((M + nchunks - 1) / nchunks);
+  auto part = torch::empty({nchunks, 2, H}, x.options().dtype(torch::kFloat32));
+  auto dwdb = torch::empty({2, H}, x.options().dtype(torch::kFloat32));
+  dim3 gridc((H + 255) / 256, nchunks), blockc(256);
   if (x.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL(layernorm_bwd_dx_kernel<__hip_bfloat16>, grid, block, 0, stream,
                        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
                        mean.data_ptr<float>(), invstd.data_ptr<float>(),
                        (const __hip_bfloat16*)dy.data_ptr(), (__hip_bfloat16*)dx.data_ptr(), H);
-    hipLaunchKernelGGL(layernorm_bwd_dwdb_kernel<__hip_bfloat16>, gridc, blockc, 0, stream,
-                       (const __hip_bfloat16*)x.data_ptr(), mean.data_ptr<float>(),
+    hipLaunchKernelGGL(layernorm_bwd_dwdb_partial_kernel<__hip_bfloat16>, gridc, blockc, 0,
+                       stream, (const __hip_bfloat16*)x.data_ptr(), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), (const __hip_bfloat16*)dy.data_ptr(),
-                       dw.data_ptr<float>(), db.data_ptr<float>(), M, H);
+                       part.data_ptr<float>(), M, H, rows_per_chunk);
   } else {
     hipLaunchKernelGGL(layernorm_bwd_dx_kernel<float>, grid, block, 0, stream,
                        x.data_ptr<float>(), w.data_ptr<float>(), mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), dy.data_ptr<float>(), dx.data_ptr<float>(), H);
-    hipLaunchKernelGGL(layernorm_bwd_dwdb_kernel<float>, gridc, blockc, 0, stream,
+    hipLaunchKernelGGL(layernorm_bwd_dwdb_partial_kernel<float>, gridc, blockc, 0, stream,
                        x.data_ptr<float>(), mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       dy.data_ptr<float>(), dw.data_ptr<float>(), db.data_ptr<float>(), M, H);
+                       dy.data_ptr<float>(), part.data_ptr<float>(), M, H, rows_per_chunk);
   }
+  hipLaunchKernelGGL(col_combine_kernel, dim3((2 * H + 255) / 256), blockc, 0, stream,
+                     part.data_ptr<float>(), dwdb.data_ptr<float>(), H, nchunks, 2);
   HIP_CHECK_LAST();
-  return {dx, dw, db};
+  return {dx, dwdb[0], dwdb[1]};
 }

@@ -56,6 +56,41 @@ def test_rmsnorm_fwd_bwd(H, dtype):
     assert_close_bf16(dw.float(), wr.grad, atol=5e-2, rtol=5e-2, what="rmsnorm dw")
 
 
+def test_rmsnorm_bwd_dw_large_m():
+    """Many-chunk path of the two-stage dw reduction (M >> chunk rows)."""
+    torch.manual_seed(1)
+    M, H = 4096 + 17, 512
+    x = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    y, invrms = ext().rmsnorm_fwd(x, w, 1e-6)
+    dy = torch.randn_like(x)
+    dx, dw = ext().rmsnorm_bwd(x, w, invrms, dy)
+    xr = x.float().requires_grad_(True)
+    wr = w.float().requires_grad_(True)
+    var = xr.pow(2).mean(-1, keepdim=True)
+    ref = wr * (xr * torch.rsqrt(var + 1e-6))
+    ref.backward(dy.float())
+    assert_close_bf16(dw.float(), wr.grad, atol=5e-1, rtol=5e-2, what="rmsnorm dw large-M")
+
+
+def test_layernorm_bwd_dwdb_large_m():
+    torch.manual_seed(2)
+    M, H = 4096 + 5, 384
+    x = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    y, mean, invstd = ext().layernorm_fwd(x, w, b, 1e-5)
+    dy = torch.randn_like(x)
+    dx, dw, db = ext().layernorm_bwd(x, w, mean, invstd, dy)
+    xr = x.float().requires_grad_(True)
+    wr = w.float().requires_grad_(True)
+    br = b.float().requires_grad_(True)
+    ref = torch.nn.functional.layer_norm(xr, (H,), wr, br, 1e-5)
+    ref.backward(dy.float())
+    assert_close_bf16(dw.float(), wr.grad, atol=5e-1, rtol=5e-2, what="ln dw large-M")
+    assert_close_bf16(db.float(), br.grad, atol=5e-1, rtol=5e-2, what="ln db large-M")
+
+
 # ---------------------------------------------------------------------------
 # LayerNorm
 # ---------------------------------------------------------------------------
