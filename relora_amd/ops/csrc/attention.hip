@@ -5,18 +5,29 @@
 // modeling_pythia.py:264-288): causal-only, no padding mask, dropout_p=0 —
 // exactly the training configuration the reference uses.
 //
-// Structure (forward): block = 4 waves = 64 q rows (16 per wave), KV tiles
-// of 32. K is staged row-major in LDS ([kv][hd], read as contiguous-k B
-// fragments), V is staged transposed ([hd][kv]). Online softmax runs fully
-// in registers on the MFMA C-layout (row r of a 16x16 tile lives in the 16
-// lanes with l>>4 == r>>2 at register r&3; row reductions are 4 shfl_xor
-// steps over the low 4 lane bits). P is redistributed to A-fragment layout
-// through a small per-wave LDS buffer. head_dim <= 128, any S; head dims
-// that are not multiples of 32 (e.g. 48) are zero-padded in the K-dim.
+// Structure (v2): 512-thread / 8-wave workgroups, 128 query (or kv) rows
+// per block (16 per wave), 64-row K/V (or Q/dO) tiles, double-buffered LDS
+// staging with register prefetch (issue the next tile's global loads before
+// computing on the current one — T14 split), vectorized row staging plus
+// vector-load/scalar-LDS-write transposes, and a per-wave private LDS
+// region for the P/dS C→A-fragment relayout so the softmax step needs no
+// block barrier.  Two __syncthreads per tile (the v1 kernel paid six per
+// 64 kv rows and ran at ~70 TF).
 //
-// Backward: standard FlashAttention-2 split — a delta preprocess
-// (rowsum(dO*O)), a dQ kernel (blocks over q tiles) and a dK/dV kernel
-// (blocks over kv tiles), each recomputing P from the saved LSE.
+// Tiles are row-padded by 8 bf16 (16 B): row stride 144 B = 36 dwords, so
+// the 16-lane ds_read_b128 groups of consecutive rows land on disjoint
+// 4-dword bank windows (36*r mod 64 for r=0..15 covers all banks) — LDS
+// reads are conflict-free without an XOR swizzle.
+//
+// Online softmax runs fully in registers on the MFMA C-layout (row r of a
+// 16x16 tile lives in the 16 consecutive lanes with l>>4 == r>>2 at
+// register r&3; row reductions are 4 shfl_xor steps over the low 4 lane
+// bits).  head_dim <= 128; dims that are not multiples of 32 are
+// zero-padded in the K-dim.
+//
+// Backward: FlashAttention-2 split — delta preprocess (rowsum(dO*O)), a dQ
+// kernel (blocks over 128 q rows) and a dK/dV kernel (blocks over 128 kv
+// rows), each recomputing P from the saved LSE.
 
 #include <ATen/hip/HIPContext.h>
 #include <torch/extension.h>
@@ -26,7 +37,8 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-#define LPAD 8  // bf16 elements of row padding in LDS tiles (16B)
+#define LPAD 8     // bf16 elements of row padding in LDS tiles (16B)
+#define TILE 64    // staged tile rows (kv rows fwd/dq, q rows dkdv)
 
 DEV_INLINE float bf_to_f(__bf16 x) { return (float)x; }
 
@@ -40,39 +52,6 @@ DEV_INLINE float rowgroup_sum(float x) {
 #pragma unroll
   for (int m = 1; m < 16; m <<= 1) x += __shfl_xor(x, m);
   return x;
-}
-
-// stage a [rows x cols_pad] tile row-major into LDS from global [S, hd],
-// zero-padding rows >= S and cols >= hd. cols_pad is a multiple of 8.
-DEV_INLINE void stage_rows(__bf16* dst, const __hip_bfloat16* src, int row0,
-                           int S, int hd, int rows, int cols_pad, int ldst) {
-  const int total = rows * cols_pad / 8;
-  for (int t = threadIdx.x; t < total; t += blockDim.x) {
-    const int r = t / (cols_pad / 8);
-    const int c = (t % (cols_pad / 8)) * 8;
-    __bf16* d = dst + r * ldst + c;
-    const int gr = row0 + r;
-    if (gr < S && c + 8 <= hd) {
-      bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long)gr * hd + c);
-      *reinterpret_cast<bf16x8*>(d) = v;
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        d[j] = (gr < S && c + j < hd) ? (__bf16)src[(long)gr * hd + c + j] : (__bf16)0.f;
-    }
-  }
-}
-
-// stage a transposed tile: dst[hd][rows] from global [S, hd]
-DEV_INLINE void stage_rows_t(__bf16* dst, const __hip_bfloat16* src, int row0,
-                             int S, int hd, int rows, int ldst) {
-  const int total = rows * hd;
-  for (int t = threadIdx.x; t < total; t += blockDim.x) {
-    const int r = t / hd;   // kv/q row
-    const int c = t % hd;   // feature
-    const int gr = row0 + r;
-    dst[c * ldst + r] = (gr < S) ? (__bf16)src[(long)gr * hd + c] : (__bf16)0.f;
-  }
 }
 
 // load an A/B fragment from an LDS tile: lane reads row `row`, 8 elements
@@ -96,27 +75,76 @@ DEV_INLINE bf16x8 global_frag(const __hip_bfloat16* src, int grow, int S, int hd
 }
 
 // ---------------------------------------------------------------------------
+// register-prefetch staging: a TILE x HD tile as NV bf16x8 slices per thread
+// (512-thread mapping: slot = tid + i*512, row = slot/(HD/8), c8 = slot%(HD/8))
+// ---------------------------------------------------------------------------
+
+template <int HD, int NV>
+DEV_INLINE void tile_load_regs(bf16x8 (&r)[NV], const __hip_bfloat16* src,
+                               int row0, int S, int hd) {
+  constexpr int C8 = HD / 8;
+#pragma unroll
+  for (int i = 0; i < NV; ++i) {
+    const int slot = threadIdx.x + i * 512;
+    if (slot >= TILE * C8) break;
+    const int row = slot / C8;
+    const int c = (slot % C8) * 8;
+    r[i] = global_frag(src, row0 + row, S, hd, c);
+  }
+}
+
+// write the registered tile row-major into LDS [TILE][ld]
+template <int HD, int NV>
+DEV_INLINE void tile_write_rows(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
+  constexpr int C8 = HD / 8;
+#pragma unroll
+  for (int i = 0; i < NV; ++i) {
+    const int slot = threadIdx.x + i * 512;
+    if (slot >= TILE * C8) break;
+    const int row = slot / C8;
+    const int c = (slot % C8) * 8;
+    *reinterpret_cast<bf16x8*>(dst + row * ld + c) = r[i];
+  }
+}
+
+// write the registered tile transposed into LDS [HD][ld]: dst[c][row]
+template <int HD, int NV>
+DEV_INLINE void tile_write_t(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
+  constexpr int C8 = HD / 8;
+#pragma unroll
+  for (int i = 0; i < NV; ++i) {
+    const int slot = threadIdx.x + i * 512;
+    if (slot >= TILE * C8) break;
+    const int row = slot / C8;
+    const int c = (slot % C8) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dst[(c + j) * ld + row] = r[i][j];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------------
 
 template <int HD>  // padded head dim (multiple of 32), actual hd passed in
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+__global__ __launch_bounds__(512) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int S, int hd, float scale) {
-  constexpr int KFRAGS = HD / 32;   // QK^T k-steps
-  constexpr int NT_HD = HD / 16;    // PV hd tiles
-  constexpr int LDK = HD + LPAD;    // lds_k row stride (elements)
-  constexpr int LDV = 32 + LPAD;    // lds_vt row stride
-  constexpr int LDP = 32 + LPAD;    // lds_p row stride
+  constexpr int KFRAGS = HD / 32;      // QK^T k-steps
+  constexpr int NT_HD = HD / 16;       // PV hd tiles
+  constexpr int LDK = HD + LPAD;
+  constexpr int LDV = TILE + LPAD;
+  constexpr int LDP = TILE + LPAD;
+  constexpr int NV = (HD + 63) / 64;   // bf16x8 staging slices per thread
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* lds_k = (__bf16*)smem;                       // [32][LDK]
-  __bf16* lds_vt = lds_k + 32 * LDK;                   // [HD][LDV]
-  __bf16* lds_p = lds_vt + HD * LDV;                   // [4][16][LDP]
+  __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK]
+  __bf16* lds_vt = lds_k + 2 * TILE * LDK;    // [2][HD][LDV]
+  __bf16* lds_p = lds_vt + 2 * HD * LDV;      // [8][16][LDP]
 
   const int bh = blockIdx.y;
-  const int q_start = blockIdx.x * 64;
+  const int q_start = blockIdx.x * 128;
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -128,8 +156,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kgrp = lane >> 4;       // k-element group (x8)
 
   // Q fragments for this wave's 16 rows (A-layout)
-  const int qrow_local = wave * 16 + col;
-  const int qrow_abs = q_start + qrow_local;
+  const int qrow_abs = q_start + wave * 16 + col;
   bf16x8 qfrag[KFRAGS];
 #pragma unroll
   for (int kf = 0; kf < KFRAGS; ++kf)
@@ -142,70 +169,104 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
   for (int t = 0; t < NT_HD; ++t) o_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int q_max_abs = min(q_start + 63, S - 1);
-  const int n_kv_tiles = (q_max_abs / 32) + 1;  // causal bound
+  const int q_max_abs = min(q_start + 127, S - 1);
+  const int n_tiles = (q_max_abs / TILE) + 1;  // causal bound
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int kv_start = kt * 32;
-    stage_rows(lds_k, kp, kv_start, S, hd, 32, HD, LDK);
-    stage_rows_t(lds_vt, vp, kv_start, S, hd, 32, LDV);
-    __syncthreads();
+  bf16x8 rk[NV], rv[NV];
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
+  tile_write_t<HD, NV>(lds_vt, rv, LDV);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
+  __syncthreads();
 
-    // scores: 2 N-subtiles of 16 kv cols
-    float p_val[2][4];
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int cur = kt & 1;
+    const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
+    const __bf16* vb = lds_vt + cur * HD * LDV;
+    // per-element causal/valid checks only where the tile crosses the
+    // diagonal or the sequence end
+    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
+
+    float p_val[4][4];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kf = 0; kf < KFRAGS; ++kf) {
-        bf16x8 b = lds_frag(lds_k, n * 16 + col, kf * 32 + kgrp * 8, LDK);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kf], b, acc, 0, 0, 0);
-      }
-      const int kv_abs = kv_start + n * 16 + col;
+      for (int kf = 0; kf < KFRAGS; ++kf)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qfrag[kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK), acc, 0, 0, 0);
+      if (edge) {
+        const int kv_abs = kv0 + n * 16 + col;
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int row_abs = q_start + wave * 16 + kgrp * 4 + reg;
-        float s = acc[reg] * scale;
-        if (kv_abs > row_abs || kv_abs >= S) s = -INFINITY;
-        p_val[n][reg] = s;
+        for (int reg = 0; reg < 4; ++reg) {
+          const int row_abs = q_start + wave * 16 + kgrp * 4 + reg;
+          float s = acc[reg] * scale;
+          if (kv_abs > row_abs || kv_abs >= S) s = -INFINITY;
+          p_val[n][reg] = s;
+        }
+      } else {
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) p_val[n][reg] = acc[reg] * scale;
       }
     }
 
     // online softmax per row (4 regs per lane)
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      float rmax = fmaxf(p_val[0][reg], p_val[1][reg]);
+      float rmax = fmaxf(fmaxf(p_val[0][reg], p_val[1][reg]),
+                         fmaxf(p_val[2][reg], p_val[3][reg]));
       rmax = rowgroup_max(rmax);
-      float m_new = fmaxf(m_run[reg], rmax);
-      float alpha = (m_new == -INFINITY) ? 1.f : __expf(m_run[reg] - m_new);
-      float p0 = (m_new == -INFINITY) ? 0.f : __expf(p_val[0][reg] - m_new);
-      float p1 = (m_new == -INFINITY) ? 0.f : __expf(p_val[1][reg] - m_new);
-      p_val[0][reg] = p0;
-      p_val[1][reg] = p1;
-      float rsum = rowgroup_sum(p0 + p1);
+      const float m_new = fmaxf(m_run[reg], rmax);
+      const float alpha = (m_new == -INFINITY) ? 1.f : __expf(m_run[reg] - m_new);
+      float rsum = 0.f;
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const float p = (m_new == -INFINITY) ? 0.f : __expf(p_val[n][reg] - m_new);
+        p_val[n][reg] = p;
+        rsum += p;
+      }
+      rsum = rowgroup_sum(rsum);
       l_run[reg] = l_run[reg] * alpha + rsum;
       m_run[reg] = m_new;
 #pragma unroll
       for (int t = 0; t < NT_HD; ++t) o_acc[t][reg] *= alpha;
     }
 
-    // redistribute P (C-layout) -> A-layout via per-wave LDS
+    // redistribute P (C-layout) -> A-layout via this wave's private LDS
+    // region: no block barrier, the compiler orders the wave's own
+    // ds_write -> ds_read dependency
     __bf16* pw = lds_p + wave * 16 * LDP;
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         pw[(kgrp * 4 + reg) * LDP + n * 16 + col] = (__bf16)p_val[n][reg];
-    __syncthreads();  // also protects lds_k/vt before restage
 
-    // PV: one K=32 step over the kv tile
-    bf16x8 a = lds_frag(pw, col, kgrp * 8, LDP);
+    // PV: two K=32 steps over the 64-row kv tile
 #pragma unroll
-    for (int t = 0; t < NT_HD; ++t) {
-      bf16x8 b = lds_frag(lds_vt, t * 16 + col, kgrp * 8, LDV);
-      o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, o_acc[t], 0, 0, 0);
+    for (int ks = 0; ks < 2; ++ks) {
+      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDP);
+#pragma unroll
+      for (int t = 0; t < NT_HD; ++t)
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, lds_frag(vb, t * 16 + col, ks * 32 + kgrp * 8, LDV), o_acc[t], 0, 0, 0);
     }
     __syncthreads();
+
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * LDV, rv, LDV);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+      __syncthreads();
+    }
   }
 
   // epilogue: O = o_acc / l, LSE = m + log(l)
@@ -242,11 +303,11 @@ __global__ void attn_delta_kernel(const __hip_bfloat16* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// backward: dQ kernel — blocks over q tiles of 64 rows
+// backward: dQ kernel — blocks over 128 q rows, kv tiles of 64
 // ---------------------------------------------------------------------------
 
 template <int HD>
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -254,16 +315,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   constexpr int KFRAGS = HD / 32;
   constexpr int NT_HD = HD / 16;
   constexpr int LDK = HD + LPAD;
-  constexpr int LDT = 32 + LPAD;
+  constexpr int LDT = TILE + LPAD;
+  constexpr int NV = (HD + 63) / 64;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* lds_k = (__bf16*)smem;            // [32][LDK]  K rows
-  __bf16* lds_v = lds_k + 32 * LDK;         // [32][LDK]  V rows
-  __bf16* lds_kt = lds_v + 32 * LDK;        // [HD][LDT]  K transposed
-  __bf16* lds_p = lds_kt + HD * LDT;        // [4][16][LDT]
+  __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK] K rows
+  __bf16* lds_v = lds_k + 2 * TILE * LDK;     // [2][TILE][LDK] V rows
+  __bf16* lds_kt = lds_v + 2 * TILE * LDK;    // [2][HD][LDT]   K transposed
+  __bf16* lds_p = lds_kt + 2 * HD * LDT;      // [8][16][LDT]
 
   const int bh = blockIdx.y;
-  const int q_start = blockIdx.x * 64;
+  const int q_start = blockIdx.x * 128;
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -295,55 +357,84 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int t = 0; t < NT_HD; ++t) dq_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int q_max_abs = min(q_start + 63, S - 1);
-  const int n_kv_tiles = (q_max_abs / 32) + 1;
+  const int q_max_abs = min(q_start + 127, S - 1);
+  const int n_tiles = (q_max_abs / TILE) + 1;
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int kv_start = kt * 32;
-    stage_rows(lds_k, kp, kv_start, S, hd, 32, HD, LDK);
-    stage_rows(lds_v, vp, kv_start, S, hd, 32, HD, LDK);
-    stage_rows_t(lds_kt, kp, kv_start, S, hd, 32, LDT);
-    __syncthreads();
+  bf16x8 rk[NV], rv[NV];
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
+  tile_write_rows<HD, NV>(lds_v, rv, LDK);
+  tile_write_t<HD, NV>(lds_kt, rk, LDT);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
+  __syncthreads();
 
-    float ds_val[2][4];
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int cur = kt & 1;
+    const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
+    const __bf16* vb = lds_v + cur * TILE * LDK;
+    const __bf16* ktb = lds_kt + cur * HD * LDT;
+    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
+
+    float ds_val[4][4];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
       f32x4 s_acc = {0.f, 0.f, 0.f, 0.f};
       f32x4 dp_acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kf = 0; kf < KFRAGS; ++kf) {
-        bf16x8 bk = lds_frag(lds_k, n * 16 + col, kf * 32 + kgrp * 8, LDK);
-        bf16x8 bv = lds_frag(lds_v, n * 16 + col, kf * 32 + kgrp * 8, LDK);
-        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kf], bk, s_acc, 0, 0, 0);
-        dp_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[kf], bv, dp_acc, 0, 0, 0);
+        s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qfrag[kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK), s_acc, 0, 0, 0);
+        dp_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dofrag[kf], lds_frag(vb, n * 16 + col, kf * 32 + kgrp * 8, LDK), dp_acc, 0, 0, 0);
       }
-      const int kv_abs = kv_start + n * 16 + col;
+      const int kv_abs = kv0 + n * 16 + col;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int row_abs = q_start + wave * 16 + kgrp * 4 + reg;
-        float p = 0.f;
-        if (kv_abs <= row_abs && kv_abs < S && row_abs < S)
+        float p;
+        if (edge) {
+          p = (kv_abs <= row_abs && kv_abs < S && row_abs < S)
+                  ? __expf(s_acc[reg] * scale - lse_r[reg]) : 0.f;
+        } else {
           p = __expf(s_acc[reg] * scale - lse_r[reg]);
+        }
         ds_val[n][reg] = p * (dp_acc[reg] - delta_r[reg]) * scale;
       }
     }
 
-    // redistribute dS -> A layout
+    // redistribute dS -> A layout (intra-wave)
     __bf16* pw = lds_p + wave * 16 * LDT;
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)ds_val[n][reg];
-    __syncthreads();
 
-    bf16x8 a = lds_frag(pw, col, kgrp * 8, LDT);
 #pragma unroll
-    for (int t = 0; t < NT_HD; ++t) {
-      bf16x8 b = lds_frag(lds_kt, t * 16 + col, kgrp * 8, LDT);
-      dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, dq_acc[t], 0, 0, 0);
+    for (int ks = 0; ks < 2; ++ks) {
+      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
+#pragma unroll
+      for (int t = 0; t < NT_HD; ++t)
+        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, lds_frag(ktb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dq_acc[t], 0, 0, 0);
     }
     __syncthreads();
+
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * LDT, rk, LDT);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+      __syncthreads();
+    }
   }
 
 #pragma unroll
@@ -360,11 +451,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// backward: dK/dV kernel — blocks over kv tiles of 64 rows
+// backward: dK/dV kernel — blocks over 128 kv rows, q tiles of 64
 // ---------------------------------------------------------------------------
 
 template <int HD>
-__global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
+__global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -373,17 +464,20 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   constexpr int KFRAGS = HD / 32;
   constexpr int NT_HD = HD / 16;
   constexpr int LDK = HD + LPAD;
-  constexpr int LDT = 32 + LPAD;
+  constexpr int LDT = TILE + LPAD;
+  constexpr int NV = (HD + 63) / 64;
+  constexpr int NBUF = (HD <= 64) ? 2 : 1;  // hd128: 4 double-buffered tiles
+                                            // exceed 160 KiB LDS
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* lds_q = (__bf16*)smem;            // [32][LDK]  Q rows
-  __bf16* lds_do = lds_q + 32 * LDK;        // [32][LDK]  dO rows
-  __bf16* lds_qt = lds_do + 32 * LDK;       // [HD][LDT]  Q transposed
-  __bf16* lds_dot = lds_qt + HD * LDT;      // [HD][LDT]  dO transposed
-  __bf16* lds_p = lds_dot + HD * LDT;       // [4][16][LDT]
+  __bf16* lds_q = (__bf16*)smem;                  // [NBUF][TILE][LDK]
+  __bf16* lds_do = lds_q + NBUF * TILE * LDK;     // [NBUF][TILE][LDK]
+  __bf16* lds_qt = lds_do + NBUF * TILE * LDK;    // [NBUF][HD][LDT]
+  __bf16* lds_dot = lds_qt + NBUF * HD * LDT;     // [NBUF][HD][LDT]
+  __bf16* lds_p = lds_dot + NBUF * HD * LDT;      // [8][16][LDT]
 
   const int bh = blockIdx.y;
-  const int kv_start_blk = blockIdx.x * 64;
+  const int kv_start_blk = blockIdx.x * 128;
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -410,77 +504,105 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     dv_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int first_qt = kv_start_blk / 32;  // causal: q >= kv
-  const int n_q_tiles = (S + 31) / 32;
+  const int first_qt = kv_start_blk / TILE;  // causal: q >= kv
+  const int n_q_tiles = (S + TILE - 1) / TILE;
+
+  bf16x8 rq[NV], rdo[NV];
+  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd);
+  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
+  tile_write_rows<HD, NV>(lds_q, rq, LDK);
+  tile_write_rows<HD, NV>(lds_do, rdo, LDK);
+  tile_write_t<HD, NV>(lds_qt, rq, LDT);
+  tile_write_t<HD, NV>(lds_dot, rdo, LDT);
+  if (first_qt + 1 < n_q_tiles) {
+    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd);
+    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd);
+  }
+  __syncthreads();
 
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
-    const int q_start = qt * 32;
-    stage_rows(lds_q, qp, q_start, S, hd, 32, HD, LDK);
-    stage_rows(lds_do, dop, q_start, S, hd, 32, HD, LDK);
-    stage_rows_t(lds_qt, qp, q_start, S, hd, 32, LDT);
-    stage_rows_t(lds_dot, dop, q_start, S, hd, 32, LDT);
-    __syncthreads();
+    const int cur = (NBUF == 2) ? (qt & 1) : 0;
+    const int q_start = qt * TILE;
+    const __bf16* qb = lds_q + cur * TILE * LDK;
+    const __bf16* dob = lds_do + cur * TILE * LDK;
+    const __bf16* qtb = lds_qt + cur * HD * LDT;
+    const __bf16* dotb = lds_dot + cur * HD * LDT;
+    const bool edge = (q_start < kv_start_blk + 127) || (q_start + TILE > S);
 
     // T = K Q^T (scores transposed), dPT = V dO^T
-    float pt_val[2][4], dst_val[2][4];
+    float pt_val[4][4], dst_val[4][4];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
       f32x4 t_acc = {0.f, 0.f, 0.f, 0.f};
       f32x4 dpt_acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kf = 0; kf < KFRAGS; ++kf) {
-        bf16x8 bq = lds_frag(lds_q, n * 16 + col, kf * 32 + kgrp * 8, LDK);
-        bf16x8 bdo = lds_frag(lds_do, n * 16 + col, kf * 32 + kgrp * 8, LDK);
-        t_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[kf], bq, t_acc, 0, 0, 0);
-        dpt_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[kf], bdo, dpt_acc, 0, 0, 0);
+        t_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            kfrag[kf], lds_frag(qb, n * 16 + col, kf * 32 + kgrp * 8, LDK), t_acc, 0, 0, 0);
+        dpt_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            vfrag[kf], lds_frag(dob, n * 16 + col, kf * 32 + kgrp * 8, LDK), dpt_acc, 0, 0, 0);
       }
       const int q_abs = q_start + n * 16 + col;
       const float lse_c = (q_abs < S) ? lse[(long)bh * S + q_abs] : 0.f;
       const float delta_c = (q_abs < S) ? delta[(long)bh * S + q_abs] : 0.f;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const int kv_abs = kv_start_blk + wave * 16 + kgrp * 4 + reg;
-        float p = 0.f;
-        if (q_abs >= kv_abs && q_abs < S && kv_abs < S)
+        float p;
+        if (edge) {
+          const int kv_abs = kv_start_blk + wave * 16 + kgrp * 4 + reg;
+          p = (q_abs >= kv_abs && q_abs < S && kv_abs < S)
+                  ? __expf(t_acc[reg] * scale - lse_c) : 0.f;
+        } else {
           p = __expf(t_acc[reg] * scale - lse_c);
+        }
         pt_val[n][reg] = p;
         dst_val[n][reg] = p * (dpt_acc[reg] - delta_c) * scale;
       }
     }
 
     __bf16* pw = lds_p + wave * 16 * LDT;
-    // pass 1: dV += P^T @ dO
+    // pass 1: dV += P^T @ dO  (intra-wave relayout, no block barrier)
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)pt_val[n][reg];
-    __syncthreads();
-    {
-      bf16x8 a = lds_frag(pw, col, kgrp * 8, LDT);
 #pragma unroll
-      for (int t = 0; t < NT_HD; ++t) {
-        bf16x8 b = lds_frag(lds_dot, t * 16 + col, kgrp * 8, LDT);
-        dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, dv_acc[t], 0, 0, 0);
-      }
+    for (int ks = 0; ks < 2; ++ks) {
+      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
+#pragma unroll
+      for (int t = 0; t < NT_HD; ++t)
+        dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, lds_frag(dotb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dv_acc[t], 0, 0, 0);
     }
-    __syncthreads();
     // pass 2: dK += dS^T @ Q
 #pragma unroll
-    for (int n = 0; n < 2; ++n)
+    for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)dst_val[n][reg];
-    __syncthreads();
-    {
-      bf16x8 a = lds_frag(pw, col, kgrp * 8, LDT);
 #pragma unroll
-      for (int t = 0; t < NT_HD; ++t) {
-        bf16x8 b = lds_frag(lds_qt, t * 16 + col, kgrp * 8, LDT);
-        dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, dk_acc[t], 0, 0, 0);
-      }
+    for (int ks = 0; ks < 2; ++ks) {
+      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
+#pragma unroll
+      for (int t = 0; t < NT_HD; ++t)
+        dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, lds_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dk_acc[t], 0, 0, 0);
     }
     __syncthreads();
+
+    if (qt + 1 < n_q_tiles) {
+      const int nxt = (NBUF == 2) ? (cur ^ 1) : 0;
+      tile_write_rows<HD, NV>(lds_q + nxt * TILE * LDK, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do + nxt * TILE * LDK, rdo, LDK);
+      tile_write_t<HD, NV>(lds_qt + nxt * HD * LDT, rq, LDT);
+      tile_write_t<HD, NV>(lds_dot + nxt * HD * LDT, rdo, LDT);
+      if (qt + 2 < n_q_tiles) {
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+      }
+      __syncthreads();
+    }
   }
 
 #pragma unroll
@@ -524,10 +646,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  dim3 grid((S + 63) / 64, B * nh), block(256);
+  dim3 grid((S + 127) / 128, B * nh), block(512);
   DISPATCH_HD(HDP, {
-    const int LDK = HD + LPAD, LDV = 32 + LPAD, LDP = 32 + LPAD;
-    size_t smem = (32 * LDK + HD * LDV + 4 * 16 * LDP) * sizeof(__bf16);
+    const int LDK = HD + LPAD, LDV = TILE + LPAD, LDP = TILE + LPAD;
+    size_t smem = (2 * TILE * LDK + 2 * HD * LDV + 8 * 16 * LDP) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
@@ -554,19 +676,21 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      delta.data_ptr<float>(), hd);
   HIP_CHECK_LAST();
 
-  dim3 block(256);
+  dim3 block(512);
   DISPATCH_HD(HDP, {
-    const int LDK = HD + LPAD, LDT = 32 + LPAD;
-    size_t smem_dq = (2 * 32 * LDK + HD * LDT + 4 * 16 * LDT) * sizeof(__bf16);
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 63) / 64, B * nh), block,
+    const int LDK = HD + LPAD, LDT = TILE + LPAD;
+    size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * LDT + 8 * 16 * LDT) * sizeof(__bf16);
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dq, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
     HIP_CHECK_LAST();
-    size_t smem_dkdv = (2 * 32 * LDK + 2 * HD * LDT + 4 * 16 * LDT) * sizeof(__bf16);
-    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 63) / 64, B * nh), block,
+    constexpr int NBUF = (HD <= 64) ? 2 : 1;
+    size_t smem_dkdv =
+        (NBUF * TILE * LDK * 2 + NBUF * HD * LDT * 2 + 8 * 16 * LDT) * sizeof(__bf16);
+    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dkdv, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),

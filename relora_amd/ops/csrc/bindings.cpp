@@ -29,6 +29,11 @@ void fused_adamw(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
                  double lr, double beta1, double beta2, double eps, double wd, long step);
 torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> grads);
 void multi_tensor_scale_(std::vector<torch::Tensor> grads, double scale);
+// lora_gemm.hip
+std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t seed);
+void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q);
+void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
+                  torch::Tensor mask, double inv_keep);
 // attention.hip
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     double scale);
@@ -49,6 +54,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (gfx950)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "multi-tensor L2 norm (gfx950)");
   m.def("multi_tensor_scale_", &multi_tensor_scale_, "multi-tensor scale (gfx950)");
+  m.def("dropout_mask_fwd", &dropout_mask_fwd, "fused dropout + packed mask (gfx950)");
+  m.def("lora_add_nt_", &lora_add_nt_, "out += P @ Q^T rank-r MFMA accumulate (gfx950)");
+  m.def("lora_add_nn_", &lora_add_nn_, "out += maskscale*(P @ Q) rank-r MFMA accumulate (gfx950)");
   m.def("attn_fwd", &attn_fwd, "causal flash attention forward (gfx950 MFMA)");
   m.def("attn_bwd", &attn_bwd, "causal flash attention backward (gfx950 MFMA)");
 }

@@ -1,0 +1,256 @@
+#include "hip/hip_runtime.h"
+// Fused LoRA rank-r update kernels (K1+K2) for gfx950.
+//
+// The reference computes the LoRA path as three separate torch ops plus an
+// add (reference relora.py:319-323): dropout(x), lora_A GEMM, lora_B GEMM,
+// scale-mul, add.  Here the rank-r (<=256) reduction runs as one MFMA
+// kernel that accumulates STRAIGHT INTO the main GEMM's output, with the
+// dropout mask generated once (philox4x32-10, packed bits) and re-applied
+// in backward — removing three [M,N]-sized memory round trips per wrapped
+// Linear per direction:
+//
+//   fwd:  y[M,N]  += t[M,r] @ Bs[N,r]^T          (lora_add, TRANSQ=false)
+//   bwd:  dx[M,K] += mask/(1-p) * (u[M,r] @ A[r,K])   (TRANSQ=true, MASK)
+//
+// Tile: 128x128 out per 256-thread block (4 waves, 64x64 per wave),
+// mfma_f32_16x16x32_bf16, whole r staged in LDS once (no K loop over
+// tiles: r <= 256).  Operand tiles are staged row-major with a 16-byte row
+// pad (the attention kernels' proven layout) and read as contiguous-k
+// bf16x8 fragments.
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define LPAD 8  // bf16 elements of LDS row padding (one 16B slot)
+
+// ---------------------------------------------------------------------------
+// philox4x32-10 — counter-based RNG for the dropout mask (regenerable, but we
+// persist packed bits: exact replay in backward with zero recompute).
+// ---------------------------------------------------------------------------
+
+DEV_INLINE void philox_round(uint32_t& c0, uint32_t& c1, uint32_t& c2, uint32_t& c3,
+                             uint32_t k0, uint32_t k1) {
+  const uint32_t M0 = 0xD2511F53u, M1 = 0xCD9E8D57u;
+  uint32_t h0 = __umulhi(M0, c0), l0 = M0 * c0;
+  uint32_t h1 = __umulhi(M1, c2), l1 = M1 * c2;
+  uint32_t n0 = h1 ^ c1 ^ k0, n1 = l1, n2 = h0 ^ c3 ^ k1, n3 = l0;
+  c0 = n0; c1 = n1; c2 = n2; c3 = n3;
+}
+
+DEV_INLINE void philox4(uint64_t seed, uint64_t idx, uint32_t out[4]) {
+  uint32_t c0 = (uint32_t)idx, c1 = (uint32_t)(idx >> 32), c2 = 0x9E3779B9u, c3 = 0xBB67AE85u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  out[0] = c0; out[1] = c1; out[2] = c2; out[3] = c3;
+}
+
+// xd = keep ? x/(1-p) : 0; mask bit j of byte [m][k8] = keep(k8*8+j).
+// One thread per 8 consecutive elements (one mask byte, one bf16x8 store).
+template <typename T>
+__global__ void dropout_mask_kernel(const T* __restrict__ x, T* __restrict__ xd,
+                                    uint8_t* __restrict__ mask, long n8,
+                                    uint64_t seed, float p, float inv_keep) {
+  const long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i8 >= n8) return;
+  uint32_t r[4];
+  philox4(seed, (uint64_t)i8, r);
+  const uint32_t thr = (uint32_t)(p * 65536.0f);
+  Vec8<T> v = load8(x + i8 * 8);
+  Vec8<T> o;
+  uint8_t m = 0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const uint32_t u16 = (r[j >> 1] >> ((j & 1) * 16)) & 0xFFFFu;
+    const bool keep = u16 >= thr;
+    m |= (uint8_t)keep << j;
+    o.v[j] = keep ? from_f32<T>(to_f32(v.v[j]) * inv_keep) : from_f32<T>(0.f);
+  }
+  store8(xd + i8 * 8, o);
+  mask[i8] = m;
+}
+
+// ---------------------------------------------------------------------------
+// rank-r accumulate kernel
+// out[M,N] (+)= P[M,r] @ Q^T          (TRANSQ=false: Q is [N,r] row-major)
+// out[M,N] (+)= maskscale * (P[M,r] @ Q)   (TRANSQ=true: Q is [r,N] row-major)
+// ---------------------------------------------------------------------------
+
+template <bool TRANSQ, bool MASK>
+__global__ __launch_bounds__(256) void lora_skinny_kernel(
+    const __hip_bfloat16* __restrict__ P, const __hip_bfloat16* __restrict__ Q,
+    __hip_bfloat16* __restrict__ out, const uint8_t* __restrict__ mask,
+    float inv_keep, long M, int N, int r, int ldq) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int ldt = r + LPAD;
+  __bf16* p_im = (__bf16*)smem;            // [128][ldt]
+  __bf16* q_im = p_im + 128 * ldt;         // [128][ldt]
+
+  const long m0 = (long)blockIdx.y * 128;
+  const int n0 = blockIdx.x * 128;
+
+  // stage P rows [128][r]
+  for (int t = threadIdx.x; t < 128 * (r / 8); t += blockDim.x) {
+    const int row = t / (r / 8);
+    const int c = (t % (r / 8)) * 8;
+    bf16x8 v;
+    if (m0 + row < M) {
+      v = *reinterpret_cast<const bf16x8*>(P + (m0 + row) * (long)r + c);
+    } else {
+      v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    *reinterpret_cast<bf16x8*>(p_im + row * ldt + c) = v;
+  }
+  // stage Q tile as q_im[n][k]
+  if (!TRANSQ) {
+    for (int t = threadIdx.x; t < 128 * (r / 8); t += blockDim.x) {
+      const int n = t / (r / 8);
+      const int c = (t % (r / 8)) * 8;
+      bf16x8 v;
+      if (n0 + n < N) {
+        v = *reinterpret_cast<const bf16x8*>(Q + (n0 + n) * (long)ldq + c);
+      } else {
+        v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      *reinterpret_cast<bf16x8*>(q_im + n * ldt + c) = v;
+    }
+  } else {
+    // Q is [r][N]: q_im[n][k] = Q[k][n0+n].  Coalesced across threads in n.
+    for (int t = threadIdx.x; t < 128 * r; t += blockDim.x) {
+      const int k = t / 128;
+      const int n = t % 128;
+      q_im[n * ldt + k] =
+          (n0 + n < N) ? (__bf16)Q[(long)k * ldq + n0 + n] : (__bf16)0.f;
+    }
+  }
+  __syncthreads();
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64;  // wave row offset in tile
+  const int wc = (wave & 1) * 64;   // wave col offset
+  const int fr = lane & 15;         // fragment row/col
+  const int kg = (lane >> 4) * 8;   // k-group offset
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int kk = 0; kk < r; kk += 32) {
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const bf16x8 a =
+          *reinterpret_cast<const bf16x8*>(p_im + (wr + mi * 16 + fr) * ldt + kk + kg);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const bf16x8 b =
+            *reinterpret_cast<const bf16x8*>(q_im + (wc + ni * 16 + fr) * ldt + kk + kg);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mi][ni], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: out[m][n] += acc (bf16 read-modify-write), optional mask scale
+  const int crow = (lane >> 4) * 4;  // C-frag rows crow..crow+3, col = fr
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long m = m0 + wr + mi * 16 + crow + j;
+        const int n = n0 + wc + ni * 16 + fr;
+        if (m < M && n < N) {
+          float v = acc[mi][ni][j];
+          if (MASK) {
+            const uint8_t mb = mask[m * (long)(N >> 3) + (n >> 3)];
+            v = (mb >> (n & 7)) & 1 ? v * inv_keep : 0.f;
+          }
+          __hip_bfloat16* o = out + m * (long)N + n;
+          *o = from_f32<__hip_bfloat16>(to_f32(*o) + v);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.numel() % 8 == 0, "dropout_mask_fwd: numel must be a multiple of 8");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "dropout_mask_fwd: bf16 only");
+  auto xd = torch::empty_like(x);
+  const long n8 = x.numel() / 8;
+  auto mask = torch::empty({n8}, x.options().dtype(torch::kUInt8));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const float inv_keep = 1.f / (1.f - (float)p);
+  hipLaunchKernelGGL(dropout_mask_kernel<__hip_bfloat16>,
+                     dim3((n8 + 255) / 256), dim3(256), 0, stream,
+                     (const __hip_bfloat16*)x.data_ptr(), (__hip_bfloat16*)xd.data_ptr(),
+                     mask.data_ptr<uint8_t>(), n8, (uint64_t)seed, (float)p, inv_keep);
+  HIP_CHECK_LAST();
+  return {xd, mask};
+}
+
+// out[M,N] += P[M,r] @ Q[N,r]^T  (forward epilogue; Q = scale*lora_B.weight)
+void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q) {
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() && P.is_contiguous() && Q.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16, "lora_add: bf16 only");
+  const long M = out.size(0);
+  const int N = out.size(1);
+  const int r = P.size(1);
+  TORCH_CHECK(P.size(0) == M && Q.size(0) == N && Q.size(1) == r);
+  TORCH_CHECK(r % 32 == 0 && r <= 256, "lora_add: r must be a multiple of 32, <=256");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const size_t lds = 2u * 128 * (r + LPAD) * sizeof(__bf16);
+  dim3 grid((N + 127) / 128, (M + 127) / 128), block(256);
+  hipLaunchKernelGGL((lora_skinny_kernel<false, false>), grid, block, lds, stream,
+                     (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)Q.data_ptr(),
+                     (__hip_bfloat16*)out.data_ptr(), nullptr, 1.f, M, N, r, r);
+  HIP_CHECK_LAST();
+}
+
+// out[M,K] += maskscale * (P[M,r] @ Q[r,K])  (backward dx epilogue; Q = lora_A.weight)
+void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
+                  torch::Tensor mask, double inv_keep) {
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() && P.is_contiguous() && Q.is_contiguous());
+  TORCH_CHECK(out.scalar_type() == torch::kBFloat16, "lora_add: bf16 only");
+  const long M = out.size(0);
+  const int N = out.size(1);
+  const int r = P.size(1);
+  TORCH_CHECK(P.size(0) == M && Q.size(0) == r && Q.size(1) == N);
+  TORCH_CHECK(r % 32 == 0 && r <= 256, "lora_add: r must be a multiple of 32, <=256");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const size_t lds = 2u * 128 * (r + LPAD) * sizeof(__bf16);
+  dim3 grid((N + 127) / 128, (M + 127) / 128), block(256);
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  if (has_mask) {
+    TORCH_CHECK(N % 8 == 0, "masked lora_add: N must be a multiple of 8");
+    TORCH_CHECK(mask.numel() == M * (long)(N / 8), "mask size mismatch");
+    hipLaunchKernelGGL((lora_skinny_kernel<true, true>), grid, block, lds, stream,
+                       (const __hip_bfloat16*)P.data_ptr(),
+                       (const __hip_bfloat16*)Q.data_ptr(),
+                       (__hip_bfloat16*)out.data_ptr(), mask.data_ptr<uint8_t>(),
+                       (float)inv_keep, M, N, r, N);
+  } else {
+    hipLaunchKernelGGL((lora_skinny_kernel<true, false>), grid, block, lds, stream,
+                       (const __hip_bfloat16*)P.data_ptr(),
+                       (const __hip_bfloat16*)Q.data_ptr(),
+                       (__hip_bfloat16*)out.data_ptr(), nullptr, 1.f, M, N, r, N);
+  }
+  HIP_CHECK_LAST();
+}

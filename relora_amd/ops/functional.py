@@ -323,13 +323,88 @@ def fused_cross_entropy(hidden, weight, labels, ignore_index=-100):
 # ---------------------------------------------------------------------------
 
 
+_lora_seed_counter = [0]
+
+
+def _next_dropout_seed():
+    _lora_seed_counter[0] += 1
+    # mix with the torch seed so runs differ when the user reseeds, while
+    # successive calls in one run are distinct and deterministic
+    return (torch.initial_seed() & 0x7FFFFFFFFFFF) ^ (_lora_seed_counter[0] * 0x9E3779B97F4A7C15)
+
+
+class _FusedLoRALinear(torch.autograd.Function):
+    """GPU path: the rank-r update accumulates into the main GEMM's output
+    via the MFMA lora_add kernels; dropout runs fused with a persisted
+    packed philox mask that backward replays exactly.  Saves the three
+    [M,N]-sized epilogue round trips per Linear of the composed path."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, lora_A, lora_B, scale, dropout_p, training):
+        in_shape = x.shape
+        K = in_shape[-1]
+        N = weight.shape[0]
+        x2d = x.contiguous().view(-1, K)
+        use_dropout = dropout_p > 0 and training
+        if use_dropout:
+            seed = _next_dropout_seed() & 0x7FFFFFFFFFFFFFFF
+            xd, mask = hip.ext().dropout_mask_fwd(x2d, dropout_p, seed)
+        else:
+            xd, mask = x2d, None
+        t_u = xd @ lora_A.t()                       # [M, r]
+        bs = lora_B * scale                         # [N, r]
+        y = F.linear(x2d, weight, bias)
+        hip.ext().lora_add_nt_(y, t_u, bs)          # y += t_u @ bs^T
+        ctx.save_for_backward(x2d, xd, mask if mask is not None else x2d.new_empty(0),
+                              t_u, weight, lora_A, lora_B)
+        ctx.scale = scale
+        ctx.dropout_p = dropout_p if use_dropout else 0.0
+        ctx.has_bias = bias is not None
+        return y.view(*in_shape[:-1], N)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, xd, mask, t_u, weight, lora_A, lora_B = ctx.saved_tensors
+        scale, p = ctx.scale, ctx.dropout_p
+        N = weight.shape[0]
+        dy2d = dy.contiguous().view(-1, N)
+
+        u_s = dy2d @ (lora_B * scale)               # [M, r] = s * dy @ B
+        dx = dy2d @ weight                          # [M, K] main path
+        hip.ext().lora_add_nn_(dx, u_s, lora_A,
+                               mask if p > 0 else mask.new_empty(0, dtype=torch.uint8),
+                               1.0 / (1.0 - p) if p > 0 else 1.0)
+
+        dA = u_s.t() @ xd                           # [r, K]
+        dB = (dy2d.t() @ t_u) * scale               # [N, r]
+        dw = dy2d.t() @ x2d if ctx.needs_input_grad[1] else None
+        dbias = dy2d.sum(0) if ctx.has_bias and ctx.needs_input_grad[2] else None
+        return (dx.view(dy.shape[:-1] + (weight.shape[1],)), dw, dbias,
+                dA, dB, None, None, None)
+
+
+def _fused_ok(x, weight, lora_A, scale, lora_only):
+    r = lora_A.shape[0]
+    return (hip.use_hip(x) and not lora_only and not torch.is_tensor(scale)
+            and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
+            and r % 32 == 0 and r <= 256
+            and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+            and os.environ.get("RELORA_AMD_LORA_PATH", "fused") != "torch")
+
+
 def lora_linear(x, weight, bias, lora_A, lora_B, scale, dropout_p=0.0,
                 training=False, lora_only=False):
-    """Composed implementation (CPU, and GPU library-GEMM path).
+    """y = x W^T (+b) + s * dropout(x) A^T B^T.
 
+    GPU bf16 path: _FusedLoRALinear (MFMA rank-r accumulate kernels).
+    Otherwise: composed torch ops (CPU path and the numerics oracle).
     `scale` may be a python float or a 0-d tensor (trainable scaling,
-    already passed through tanh by the caller).
+    already passed through tanh by the caller) — tensor scale uses the
+    composed path so autograd reaches it.
     """
+    if _fused_ok(x, weight, lora_A, scale, lora_only):
+        return _FusedLoRALinear.apply(x, weight, bias, lora_A, lora_B,
+                                      float(scale), dropout_p, training)
     xd = F.dropout(x, p=dropout_p, training=training) if dropout_p > 0 else x
     lora_out = F.linear(F.linear(xd, lora_A), lora_B)
     if lora_only:

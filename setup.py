@@ -36,6 +36,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "ce.hip"),
         os.path.join(CSRC, "adamw.hip"),
         os.path.join(CSRC, "attention.hip"),
+        os.path.join(CSRC, "lora_gemm.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
