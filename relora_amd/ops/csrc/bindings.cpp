@@ -34,6 +34,12 @@ std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t s
 void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q);
 void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
                   torch::Tensor mask, double inv_keep);
+torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X);
+// quantize.hip
+std::vector<torch::Tensor> quantize_nf4(torch::Tensor x);
+torch::Tensor dequantize_nf4(torch::Tensor q, torch::Tensor absmax, long n, torch::ScalarType dtype);
+std::vector<torch::Tensor> quantize_int8(torch::Tensor x);
+torch::Tensor dequantize_int8(torch::Tensor q, torch::Tensor absmax, long n, torch::ScalarType dtype);
 // attention.hip
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     double scale);
@@ -57,6 +63,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_mask_fwd", &dropout_mask_fwd, "fused dropout + packed mask (gfx950)");
   m.def("lora_add_nt_", &lora_add_nt_, "out += P @ Q^T rank-r MFMA accumulate (gfx950)");
   m.def("lora_add_nn_", &lora_add_nn_, "out += maskscale*(P @ Q) rank-r MFMA accumulate (gfx950)");
+  m.def("skinny_grad", &skinny_grad, "P^T @ X chunked fp32 reduction (gfx950)");
+  m.def("quantize_nf4", &quantize_nf4, "NF4 blockwise quantize (gfx950)");
+  m.def("dequantize_nf4", &dequantize_nf4, "NF4 blockwise dequantize (gfx950)");
+  m.def("quantize_int8", &quantize_int8, "int8 blockwise quantize (gfx950)");
+  m.def("dequantize_int8", &dequantize_int8, "int8 blockwise dequantize (gfx950)");
   m.def("attn_fwd", &attn_fwd, "causal flash attention forward (gfx950 MFMA)");
   m.def("attn_bwd", &attn_bwd, "causal flash attention backward (gfx950 MFMA)");
 }

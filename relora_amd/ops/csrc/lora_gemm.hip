@@ -254,3 +254,136 @@ void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
   }
   HIP_CHECK_LAST();
 }
+
+// ---------------------------------------------------------------------------
+// LoRA weight gradients: out[r, C] = P[M, r]^T @ X[M, C], fp32 partials per
+// M-chunk (deterministic tree sum in the caller).  hipBLASLt runs these
+// skinny-output reductions as 32-workgroup launches (12% of the chip);
+// chunking M restores full occupancy.
+//   dA          = skinny_grad(P=u_s,  X=xd) -> [r, K]
+//   dB^T        = skinny_grad(P=t_u,  X=dy) -> [r, N]
+// ---------------------------------------------------------------------------
+
+// grid: (C/128, MCHUNKS); block 256 (4 waves); out tile [r<=128][128]
+__global__ __launch_bounds__(256) void skinny_grad_kernel(
+    const __hip_bfloat16* __restrict__ P, const __hip_bfloat16* __restrict__ X,
+    float* __restrict__ part, long M, int C, int r, int rows_per_chunk) {
+  constexpr int LDT = 64 + LPAD;  // transposed tile row stride (m dim)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* pt = (__bf16*)smem;        // [r][LDT]   P^T tile (m contiguous)
+  __bf16* xt = pt + 128 * LDT;       // [128][LDT] X^T tile (c rows, m cols)
+
+  const int c0 = blockIdx.x * 128;
+  const long m_begin = (long)blockIdx.y * rows_per_chunk;
+  const long m_end = min(M, m_begin + rows_per_chunk);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+
+  // per-wave output rows: wave*32 .. +32 (2 row frags), cols c0..c0+128
+  f32x4 acc[2][8];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (long m0 = m_begin; m0 < m_end; m0 += 64) {
+    // stage P^T: pt[j][mm] = P[m0+mm][j]; threads cover 64 rows x r/8 vec8
+    for (int t = threadIdx.x; t < 64 * (r / 8); t += blockDim.x) {
+      const int mm = t / (r / 8);
+      const int j8 = (t % (r / 8)) * 8;
+      bf16x8 v;
+      if (m0 + mm < m_end) {
+        v = *reinterpret_cast<const bf16x8*>(P + (m0 + mm) * (long)r + j8);
+      } else {
+        v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) pt[(j8 + j) * LDT + mm] = v[j];
+    }
+    // stage X^T: xt[cc][mm] = X[m0+mm][c0+cc]; vec8 over the row direction
+    for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
+      const int mm = t / 16;
+      const int c8 = (t % 16) * 8;
+      bf16x8 v;
+      if (m0 + mm < m_end && c0 + c8 + 8 <= C) {
+        v = *reinterpret_cast<const bf16x8*>(X + (m0 + mm) * (long)C + c0 + c8);
+      } else if (m0 + mm < m_end) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (c0 + c8 + j < C) ? (__bf16)X[(m0 + mm) * (long)C + c0 + c8 + j]
+                                   : (__bf16)0.f;
+      } else {
+        v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) xt[(c8 + j) * LDT + mm] = v[j];
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int jrow = wave * 32 + i * 16 + col;
+        const bf16x8 a = (jrow < r)
+            ? *reinterpret_cast<const bf16x8*>(pt + jrow * LDT + ks * 32 + kgrp * 8)
+            : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const bf16x8 b = *reinterpret_cast<const bf16x8*>(
+              xt + (j * 16 + col) * LDT + ks * 32 + kgrp * 8);
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i][j], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: fp32 partial plane for this chunk
+  float* plane = part + (long)blockIdx.y * r * C;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int rbase = wave * 32 + i * 16 + (kgrp << 2);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int c = c0 + j * 16 + col;
+      if (c >= C) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int rr = rbase + reg;
+        if (rr < r) plane[(long)rr * C + c] = acc[i][j][reg];
+      }
+    }
+  }
+}
+
+// out[r, C] fp32 = P[M,r]^T @ X[M,C] as MCHUNK partial planes summed by torch
+torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X) {
+  TORCH_CHECK(P.is_cuda() && P.is_contiguous() && X.is_contiguous());
+  TORCH_CHECK(P.scalar_type() == torch::kBFloat16 && X.scalar_type() == torch::kBFloat16);
+  const long M = P.size(0);
+  const int r = P.size(1);
+  const int C = X.size(1);
+  TORCH_CHECK(X.size(0) == M && r % 8 == 0 && r <= 128);
+  // chunk M so the grid fills the chip: (C/128)*chunks >= ~512
+  int chunks = 1;
+  const int ctiles = (C + 127) / 128;
+  while (chunks < 32 && ctiles * chunks < 512 && (M + chunks - 1) / chunks > 256)
+    chunks <<= 1;
+  int rows = (int)((M + chunks - 1) / chunks);
+  rows = (rows + 63) / 64 * 64;  // multiple of the m-tile
+  chunks = (int)((M + rows - 1) / rows);
+  auto part = torch::empty({chunks, r, C}, P.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  constexpr int LDT = 64 + LPAD;
+  const size_t lds = (128 * LDT + 128 * LDT) * sizeof(__bf16);
+  dim3 grid(ctiles, chunks), block(256);
+  hipLaunchKernelGGL(skinny_grad_kernel, grid, block, lds, stream,
+                     (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
+                     part.data_ptr<float>(), M, C, r, rows);
+  HIP_CHECK_LAST();
+  return part.sum(0);
+}

@@ -356,7 +356,7 @@ class _FusedLoRALinear(torch.autograd.Function):
         y = F.linear(x2d, weight, bias)
         hip.ext().lora_add_nt_(y, t_u, bs)          # y += t_u @ bs^T
         ctx.save_for_backward(x2d, xd, mask if mask is not None else x2d.new_empty(0),
-                              t_u, weight, lora_A, lora_B)
+                              t_u, weight, lora_A, bs)
         ctx.scale = scale
         ctx.dropout_p = dropout_p if use_dropout else 0.0
         ctx.has_bias = bias is not None
@@ -364,19 +364,21 @@ class _FusedLoRALinear(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x2d, xd, mask, t_u, weight, lora_A, lora_B = ctx.saved_tensors
+        x2d, xd, mask, t_u, weight, lora_A, bs = ctx.saved_tensors
         scale, p = ctx.scale, ctx.dropout_p
         N = weight.shape[0]
         dy2d = dy.contiguous().view(-1, N)
 
-        u_s = dy2d @ (lora_B * scale)               # [M, r] = s * dy @ B
+        u_s = dy2d @ bs                             # [M, r] = s * dy @ B
         dx = dy2d @ weight                          # [M, K] main path
         hip.ext().lora_add_nn_(dx, u_s, lora_A,
                                mask if p > 0 else mask.new_empty(0, dtype=torch.uint8),
                                1.0 / (1.0 - p) if p > 0 else 1.0)
 
-        dA = u_s.t() @ xd                           # [r, K]
-        dB = (dy2d.t() @ t_u) * scale               # [N, r]
+        # dA = u_s^T @ xd, dB = s * dy^T @ t_u = (t_u^T @ dy)^T * s — via the
+        # chunked-M skinny_grad kernel (hipBLASLt launches these 32-WG wide)
+        dA = hip.ext().skinny_grad(u_s, xd).to(lora_A.dtype)
+        dB = hip.ext().skinny_grad(t_u, dy2d).t().contiguous().mul_(scale).to(bs.dtype)
         dw = dy2d.t() @ x2d if ctx.needs_input_grad[1] else None
         dbias = dy2d.sum(0) if ctx.has_bias and ctx.needs_input_grad[2] else None
         return (dx.view(dy.shape[:-1] + (weight.shape[1],)), dw, dbias,

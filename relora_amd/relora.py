@@ -13,9 +13,10 @@ MI355X notes:
 * merge `W += B·A·scale` runs in the model dtype, exactly like the reference
   (bf16 accumulation per cycle — documented deviation point in SURVEY.md §7
   hard-part 4);
-* the 4-bit/8-bit quantized frozen-W path of the reference uses bitsandbytes
-  (CUDA-only, not in this image). `quantize=` is accepted for CLI parity and
-  raises NotImplementedError when set.
+* the 4-bit/8-bit quantized frozen-W path replaces bitsandbytes with our
+  blockwise NF4/int8 HIP kernels (relora_amd/ops/quant.py,
+  ops/csrc/quantize.hip): W lives quantized at rest, dequantizes per GEMM,
+  and the merge runs dequant -> += BA·s -> requant.
 """
 
 import json
@@ -49,11 +50,13 @@ def merge_and_reinit_functional(module):
     """Out-of-class merge (kept for FSDP-style use, reference relora.py:31-46)."""
     if not isinstance(module, ReLoRaLinear):
         return
-    if module.quantize is not None:
-        raise NotImplementedError("functional merge is not implemented for quantized layers")
     delta = module.lora_B.weight @ module.lora_A.weight
     delta = delta * module._post_lora_scale()
-    module.weight.data += delta
+    if module.quantize is None:
+        module.weight.data += delta
+    else:
+        w = module.weight.materialize(module.lora_A.weight.dtype)
+        module.weight.requantize_(w + delta)
     nn.init.kaiming_uniform_(module.lora_A.weight, a=math.sqrt(5))
     nn.init.zeros_(module.lora_B.weight)
     if module.trainable_scaling:
@@ -211,11 +214,8 @@ class ReLoRaLinear(nn.Module):
         super().__init__()
         if r <= 0:
             raise ValueError("r must be positive. If you want r == 0, use the original model.")
-        if quantize is not None:
-            raise NotImplementedError(
-                "quantized frozen weights (bitsandbytes NF4/int8) are not available "
-                "in the ROCm build yet; run without --quantize"
-            )
+        if quantize not in (None, "4bit", "8bit"):
+            raise ValueError(f"quantize must be None, '4bit' or '8bit', got {quantize!r}")
 
         if lora_only:
             self.weight = None
@@ -231,7 +231,15 @@ class ReLoRaLinear(nn.Module):
                 weight_data = torch.zeros(
                     out_features, in_features, device=device, dtype=dtype, requires_grad=False
                 )
-            self.weight = nn.Parameter(weight_data, requires_grad=False)
+            if quantize is None:
+                self.weight = nn.Parameter(weight_data, requires_grad=False)
+            else:
+                # frozen W lives blockwise-quantized (NF4 / int8) and is
+                # dequantized per GEMM (reference's bitsandbytes flow,
+                # relora.py:225-238 — ours is relora_amd/ops/quant.py)
+                from relora_amd.ops.quant import QuantizedWeight
+
+                self.weight = QuantizedWeight(weight_data, quantize)
 
         self.in_features = in_features
         self.out_features = out_features
@@ -251,7 +259,7 @@ class ReLoRaLinear(nn.Module):
             self.scaling = nn.Parameter(torch.tensor([1.0]), requires_grad=True)
         else:
             self.scaling = self.lora_alpha / self.r
-        if not self.lora_only:
+        if not self.lora_only and isinstance(self.weight, nn.Parameter):
             self.weight.requires_grad = False
 
     def _post_lora_scale(self):
@@ -259,12 +267,24 @@ class ReLoRaLinear(nn.Module):
             return self.scaling.tanh()
         return self.scaling
 
+    def _dense_weight(self):
+        """The frozen W as a dense tensor (dequantized when quantize is set)."""
+        if self.quantize is None:
+            return self.weight
+        return self.weight.materialize(self.lora_A.weight.dtype)
+
     @torch.no_grad()
     def merge_and_reinit(self):
         if self.lora_only:
             logger.warning("Skipping merge and reinit, because only lora parameters are used")
             return
-        self.weight.data += self.lora_B.weight @ self.lora_A.weight * self._post_lora_scale()
+        if self.quantize is None:
+            self.weight.data += self.lora_B.weight @ self.lora_A.weight * self._post_lora_scale()
+        else:
+            # dequant -> merge -> requant (reference relora.py:277-299)
+            w = self.weight.materialize(self.lora_A.weight.dtype)
+            w += self.lora_B.weight @ self.lora_A.weight * self._post_lora_scale()
+            self.weight.requantize_(w)
         nn.init.kaiming_uniform_(self.lora_A.weight, a=math.sqrt(5))
         nn.init.zeros_(self.lora_B.weight)
         if self.trainable_scaling:
@@ -273,7 +293,7 @@ class ReLoRaLinear(nn.Module):
     def forward(self, x: torch.Tensor):
         return ops.lora_linear(
             x,
-            self.weight,
+            self._dense_weight(),
             self.bias,
             self.lora_A.weight,
             self.lora_B.weight,

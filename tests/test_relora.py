@@ -128,10 +128,54 @@ def test_r_zero_raises(tiny_llama_config):
         ReLoRaModel(model, r=0, target_modules=["attn"])
 
 
-def test_quantize_raises(tiny_llama_config):
+def test_quantized_wrap_forward_and_merge(tiny_llama_config):
+    """4-bit frozen W: wrap, forward, merge, forward again (CPU ref path)."""
+    torch.manual_seed(0)
     model = LlamaForCausalLM(tiny_llama_config)
-    with pytest.raises(NotImplementedError):
-        ReLoRaModel(model, r=8, target_modules=["attn"], quantize="4bit")
+    wrapped = ReLoRaModel(model, r=8, lora_alpha=16, lora_dropout=0.0,
+                          target_modules=["attn", "mlp"], keep_original_weights=True,
+                          quantize="4bit")
+    x = torch.randint(0, tiny_llama_config.vocab_size, (1, 16))
+    out = wrapped(input_ids=x, labels=x)
+    assert torch.isfinite(out.loss)
+    # quantized state: no dense weight parameter on wrapped linears
+    from relora_amd.relora import ReLoRaLinear
+    lin = [m for m in wrapped.modules() if isinstance(m, ReLoRaLinear)][0]
+    assert lin.weight.qdata.dtype == torch.uint8
+    w_before = lin._dense_weight().clone()
+    with torch.no_grad():
+        lin.lora_A.weight.normal_()
+        lin.lora_B.weight.normal_()
+    delta = (lin.lora_B.weight @ lin.lora_A.weight * lin._post_lora_scale()).float()
+    wrapped.merge_and_reinit()
+    w_after = lin._dense_weight()
+    # merge landed (up to 4-bit requantization error of the merged weight)
+    target = w_before.float() + delta
+    scale_err = (w_after.float() - target).abs().max()
+    qstep = target.abs().max() * 0.2  # NF4 worst-case relative step
+    assert scale_err < qstep, (scale_err, qstep)
+    assert (lin.lora_B.weight == 0).all()
+
+
+def test_quantize_roundtrip_refs():
+    from relora_amd.ops.quant import (dequantize_int8_ref, dequantize_nf4_ref,
+                                      quantize_int8_ref, quantize_nf4_ref)
+
+    torch.manual_seed(1)
+    x = torch.randn(4096) * 0.3
+    q, am = quantize_nf4_ref(x)
+    back = dequantize_nf4_ref(q, am, 4096)
+    # NF4 relative error within each block is bounded by half the largest
+    # codebook gap (|-1.0 - -0.696| / 2 ~ 0.152) times absmax
+    blocks = x.view(-1, 64)
+    bmax = blocks.abs().amax(dim=1, keepdim=True)
+    assert ((back.view(-1, 64) - blocks).abs() / bmax.clamp_min(1e-6)).max() < 0.16
+
+    q8, am8 = quantize_int8_ref(x)
+    back8 = dequantize_int8_ref(q8, am8, 4096)
+    b8 = x.view(-1, 256)
+    m8 = b8.abs().amax(dim=1, keepdim=True)
+    assert ((back8.view(-1, 256) - b8).abs() / m8.clamp_min(1e-6)).max() < 1.0 / 127
 
 
 # ---------------------------------------------------------------------------
