@@ -195,7 +195,9 @@ def test_quantize_kernels_match_refs():
     assert torch.allclose(am8_gpu.cpu(), am8_ref, atol=1e-3)
     back8_gpu = ext().dequantize_int8(q8_gpu, am8_gpu, x.numel(), torch.float32).cpu()
     back8_ref = dequantize_int8_ref(q8_ref, am8_ref, x.numel())
-    assert (back8_gpu - back8_ref).abs().max() < 5e-3
+    # allow one quantization step for round-half ties at code boundaries
+    step = am8_ref.max().item() / 127
+    assert (back8_gpu - back8_ref).abs().max() <= step * 1.01
 
 
 def test_quantized_relora_linear_gpu():
