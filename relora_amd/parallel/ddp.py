@@ -67,6 +67,10 @@ class DistributedModel(nn.Module):
         if bucket_cap_mb is None:
             bucket_cap_mb = float(os.environ.get("RELORA_AMD_BUCKET_MB", "100"))
         self.bucket_cap_bytes = int(bucket_cap_mb * 1024 * 1024)
+        # RCCL supports in-collective averaging; gloo (CPU tests) does not
+        self._reduce_op = (dist.ReduceOp.AVG if _dist_active()
+                           and dist.get_backend(process_group) == "nccl"
+                           else dist.ReduceOp.SUM)
         # when False (non-boundary micro-steps) grads only accumulate locally
         self.require_backward_grad_sync = True
 
@@ -119,7 +123,7 @@ class DistributedModel(nn.Module):
         b.pending -= 1
         if b.pending == 0:
             b.handle = dist.all_reduce(
-                b.buffer, op=dist.ReduceOp.SUM, group=self.process_group, async_op=True
+                b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
             )
 
     # -- public API ---------------------------------------------------------
@@ -141,20 +145,21 @@ class DistributedModel(nn.Module):
                     and b.pending < len(b.params):
                 # partial bucket (should not happen in normal training)
                 b.handle = dist.all_reduce(
-                    b.buffer, op=dist.ReduceOp.SUM, group=self.process_group, async_op=True
+                    b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
                 )
             if b.handle is None and self.require_backward_grad_sync:
                 # bucket never fired (e.g. unused params): reduce it so ranks agree
                 b.handle = dist.all_reduce(
-                    b.buffer, op=dist.ReduceOp.SUM, group=self.process_group, async_op=True
+                    b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
                 )
         for b in self._buckets:
             if b.handle is not None:
                 b.handle.wait()
                 b.handle = None
-        inv = 1.0 / self.world_size
-        for b in self._buckets:
-            b.buffer.mul_(inv)
+        if self._reduce_op != dist.ReduceOp.AVG:
+            inv = 1.0 / self.world_size
+            for b in self._buckets:
+                b.buffer.mul_(inv)
         self._reset_pending()
 
     def zero_grad_buffers(self):
