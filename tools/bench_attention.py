@@ -7,7 +7,11 @@ flagship shape, plus optional SDPA comparison.
 """
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 
 import torch
 
