@@ -238,7 +238,10 @@ def flash_attention(q, k, v, causal=True, dropout_p=0.0, scale=None):
 # in-place softmax-minus-onehot gradient kernel in backward.
 # ---------------------------------------------------------------------------
 
-_CE_CHUNK = int(os.environ.get("RELORA_AMD_CE_CHUNK", "8192"))
+# 288 GB HBM comfortably holds one [M,V] bf16 logits buffer for the flagship
+# shapes (16384 x 50304 = 1.6 GB), so default to a single chunk: fewer, larger
+# GEMMs and no fp32 dw accumulation pass. Shrink via env on smaller cards.
+_CE_CHUNK = int(os.environ.get("RELORA_AMD_CE_CHUNK", "16384"))
 
 
 def _row_stats_torch(logits, labels, ignore_index):
@@ -284,7 +287,9 @@ class _FusedCrossEntropy(torch.autograd.Function):
         M, H = hidden.shape
         use_hip = hip.use_hip(hidden)
         dh = torch.empty_like(hidden)
-        dw_acc = torch.zeros(weight.shape, dtype=torch.float32, device=weight.device)
+        single_chunk = M <= _CE_CHUNK
+        dw_acc = None if single_chunk else torch.zeros(
+            weight.shape, dtype=torch.float32, device=weight.device)
         gscale = (grad_out.float() / ctx.n_valid).item() if grad_out.dim() == 0 else None
         for s in range(0, M, _CE_CHUNK):
             e = min(s + _CE_CHUNK, M)
@@ -303,8 +308,13 @@ class _FusedCrossEntropy(torch.autograd.Function):
                 p[vmask, safe[vmask]] -= 1.0
                 dlogits = (p * gscale).to(logits.dtype)
             dh[s:e] = dlogits @ weight
-            dw_acc += (dlogits.t() @ hidden[s:e]).float()
-        return dh, dw_acc.to(weight.dtype), None, None
+            if single_chunk:
+                dw = (dlogits.t() @ hidden[s:e]).to(weight.dtype)
+            else:
+                dw_acc += (dlogits.t() @ hidden[s:e]).float()
+        if not single_chunk:
+            dw = dw_acc.to(weight.dtype)
+        return dh, dw, None, None
 
 
 def fused_cross_entropy(hidden, weight, labels, ignore_index=-100):
