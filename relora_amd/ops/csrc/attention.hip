@@ -14,10 +14,12 @@
 // block barrier.  Two __syncthreads per tile (the v1 kernel paid six per
 // 64 kv rows and ran at ~70 TF).
 //
-// Tiles are row-padded by 8 bf16 (16 B): row stride 144 B = 36 dwords, so
-// the 16-lane ds_read_b128 groups of consecutive rows land on disjoint
-// 4-dword bank windows (36*r mod 64 for r=0..15 covers all banks) — LDS
-// reads are conflict-free without an XOR swizzle.
+// Row-major tiles are padded by 8 bf16 (16 B): row stride 144 B = 36
+// dwords, so 16-lane ds_read_b128 groups of consecutive rows land on
+// disjoint 4-dword bank windows (36*r mod 64 covers all banks) — reads are
+// conflict-free.  Transposed tiles (V^T, K^T, Q^T, dO^T) use the rotated
+// layout of t_rot() below: conflict-free on both the transpose writes and
+// the fragment reads.
 //
 // Online softmax runs fully in registers on the MFMA C-layout (row r of a
 // 16x16 tile lives in the 16 consecutive lanes with l>>4 == r>>2 at
@@ -107,18 +109,36 @@ DEV_INLINE void tile_write_rows(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
   }
 }
 
-// write the registered tile transposed into LDS [HD][ld]: dst[c][row]
+// Transposed tiles use a rotated layout: element (c, kv) lives at
+// c*64 + (((kv>>3) + (c>>3) + (c&7)) & 7)*8 + (kv&7).  Found by exhaustive
+// search: both the 8-scalar-per-thread transpose WRITES and the bf16x8
+// fragment READS are bank-conflict-free (the naive dst[c][kv] layout put
+// every column write of one instruction on one bank - 8-way, 16% of wave
+// cycles in the dkdv kernel).  No row padding needed: stride is exactly 64.
+DEV_INLINE int t_rot(int kv_grp, int c) {
+  return ((kv_grp + (c >> 3) + (c & 7)) & 7);
+}
+
+// read a transposed-tile B fragment: channel row c, 8 kv at kv0 (mult of 8)
+DEV_INLINE bf16x8 ldsT_frag(const __bf16* tile, int c, int kv0) {
+  return *reinterpret_cast<const bf16x8*>(tile + c * TILE + t_rot(kv0 >> 3, c) * 8);
+}
+
+// write the registered tile transposed into LDS [HD][TILE] (rotated layout)
 template <int HD, int NV>
-DEV_INLINE void tile_write_t(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
+DEV_INLINE void tile_write_t(__bf16* dst, const bf16x8 (&r)[NV]) {
   constexpr int C8 = HD / 8;
 #pragma unroll
   for (int i = 0; i < NV; ++i) {
     const int slot = threadIdx.x + i * 512;
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
-    const int c = (slot % C8) * 8;
+    const int cb = (slot % C8) * 8;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) dst[(c + j) * ld + row] = r[i][j];
+    for (int j = 0; j < 8; ++j) {
+      const int c = cb + j;
+      dst[c * TILE + t_rot(row >> 3, c) * 8 + (row & 7)] = r[i][j];
+    }
   }
 }
 
@@ -134,14 +154,13 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   constexpr int KFRAGS = HD / 32;      // QK^T k-steps
   constexpr int NT_HD = HD / 16;       // PV hd tiles
   constexpr int LDK = HD + LPAD;
-  constexpr int LDV = TILE + LPAD;
   constexpr int LDP = TILE + LPAD;
   constexpr int NV = (HD + 63) / 64;   // bf16x8 staging slices per thread
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK]
-  __bf16* lds_vt = lds_k + 2 * TILE * LDK;    // [2][HD][LDV]
-  __bf16* lds_p = lds_vt + 2 * HD * LDV;      // [8][16][LDP]
+  __bf16* lds_vt = lds_k + 2 * TILE * LDK;    // [2][HD][TILE] rotated
+  __bf16* lds_p = lds_vt + 2 * HD * TILE;     // [8][16][LDP]
 
   const int bh = blockIdx.y;
   const int q_start = blockIdx.x * 128;
@@ -176,7 +195,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
   tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
-  tile_write_t<HD, NV>(lds_vt, rv, LDV);
+  tile_write_t<HD, NV>(lds_vt, rv);
   if (n_tiles > 1) {
     tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
     tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
@@ -187,7 +206,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     const int cur = kt & 1;
     const int kv0 = kt * TILE;
     const __bf16* kb = lds_k + cur * TILE * LDK;
-    const __bf16* vb = lds_vt + cur * HD * LDV;
+    const __bf16* vb = lds_vt + cur * HD * TILE;
     // per-element causal/valid checks only where the tile crosses the
     // diagonal or the sequence end
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
@@ -254,13 +273,13 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
 #pragma unroll
       for (int t = 0; t < NT_HD; ++t)
         o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, lds_frag(vb, t * 16 + col, ks * 32 + kgrp * 8, LDV), o_acc[t], 0, 0, 0);
+            a, ldsT_frag(vb, t * 16 + col, ks * 32 + kgrp * 8), o_acc[t], 0, 0, 0);
     }
     __syncthreads();
 
     if (kt + 1 < n_tiles) {
       tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * LDV, rv, LDV);
+      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
       if (kt + 2 < n_tiles) {
         tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
         tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
@@ -321,8 +340,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK] K rows
   __bf16* lds_v = lds_k + 2 * TILE * LDK;     // [2][TILE][LDK] V rows
-  __bf16* lds_kt = lds_v + 2 * TILE * LDK;    // [2][HD][LDT]   K transposed
-  __bf16* lds_p = lds_kt + 2 * HD * LDT;      // [8][16][LDT]
+  __bf16* lds_kt = lds_v + 2 * TILE * LDK;    // [2][HD][TILE]  K transposed (rotated)
+  __bf16* lds_p = lds_kt + 2 * HD * TILE;     // [8][16][LDT]
 
   const int bh = blockIdx.y;
   const int q_start = blockIdx.x * 128;
@@ -365,7 +384,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_rows<HD, NV>(lds_v, rv, LDK);
-  tile_write_t<HD, NV>(lds_kt, rk, LDT);
+  tile_write_t<HD, NV>(lds_kt, rk);
   if (n_tiles > 1) {
     tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
     tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
@@ -377,7 +396,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const int kv0 = kt * TILE;
     const __bf16* kb = lds_k + cur * TILE * LDK;
     const __bf16* vb = lds_v + cur * TILE * LDK;
-    const __bf16* ktb = lds_kt + cur * HD * LDT;
+    const __bf16* ktb = lds_kt + cur * HD * TILE;
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
     float ds_val[4][4];
@@ -421,14 +440,14 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int t = 0; t < NT_HD; ++t)
         dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, lds_frag(ktb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dq_acc[t], 0, 0, 0);
+            a, ldsT_frag(ktb, t * 16 + col, ks * 32 + kgrp * 8), dq_acc[t], 0, 0, 0);
     }
     __syncthreads();
 
     if (kt + 1 < n_tiles) {
       tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
       tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
-      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * LDT, rk, LDT);
+      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
       if (kt + 2 < n_tiles) {
         tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
         tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
@@ -472,9 +491,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_q = (__bf16*)smem;                  // [NBUF][TILE][LDK]
   __bf16* lds_do = lds_q + NBUF * TILE * LDK;     // [NBUF][TILE][LDK]
-  __bf16* lds_qt = lds_do + NBUF * TILE * LDK;    // [NBUF][HD][LDT]
-  __bf16* lds_dot = lds_qt + NBUF * HD * LDT;     // [NBUF][HD][LDT]
-  __bf16* lds_p = lds_dot + NBUF * HD * LDT;      // [8][16][LDT]
+  __bf16* lds_qt = lds_do + NBUF * TILE * LDK;    // [NBUF][HD][TILE] rotated
+  __bf16* lds_dot = lds_qt + NBUF * HD * TILE;    // [NBUF][HD][TILE] rotated
+  __bf16* lds_p = lds_dot + NBUF * HD * TILE;     // [8][16][LDT]
 
   const int bh = blockIdx.y;
   const int kv_start_blk = blockIdx.x * 128;
@@ -512,8 +531,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
   tile_write_rows<HD, NV>(lds_q, rq, LDK);
   tile_write_rows<HD, NV>(lds_do, rdo, LDK);
-  tile_write_t<HD, NV>(lds_qt, rq, LDT);
-  tile_write_t<HD, NV>(lds_dot, rdo, LDT);
+  tile_write_t<HD, NV>(lds_qt, rq);
+  tile_write_t<HD, NV>(lds_dot, rdo);
   if (first_qt + 1 < n_q_tiles) {
     tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd);
     tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd);
@@ -525,8 +544,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const int q_start = qt * TILE;
     const __bf16* qb = lds_q + cur * TILE * LDK;
     const __bf16* dob = lds_do + cur * TILE * LDK;
-    const __bf16* qtb = lds_qt + cur * HD * LDT;
-    const __bf16* dotb = lds_dot + cur * HD * LDT;
+    const __bf16* qtb = lds_qt + cur * HD * TILE;
+    const __bf16* dotb = lds_dot + cur * HD * TILE;
     const bool edge = (q_start < kv_start_blk + 127) || (q_start + TILE > S);
 
     // T = K Q^T (scores transposed), dPT = V dO^T
@@ -573,7 +592,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
       for (int t = 0; t < NT_HD; ++t)
         dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, lds_frag(dotb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dv_acc[t], 0, 0, 0);
+            a, ldsT_frag(dotb, t * 16 + col, ks * 32 + kgrp * 8), dv_acc[t], 0, 0, 0);
     }
     // pass 2: dK += dS^T @ Q
 #pragma unroll
@@ -587,7 +606,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
       for (int t = 0; t < NT_HD; ++t)
         dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, lds_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8, LDT), dk_acc[t], 0, 0, 0);
+            a, ldsT_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8), dk_acc[t], 0, 0, 0);
     }
     __syncthreads();
 
@@ -595,8 +614,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       const int nxt = (NBUF == 2) ? (cur ^ 1) : 0;
       tile_write_rows<HD, NV>(lds_q + nxt * TILE * LDK, rq, LDK);
       tile_write_rows<HD, NV>(lds_do + nxt * TILE * LDK, rdo, LDK);
-      tile_write_t<HD, NV>(lds_qt + nxt * HD * LDT, rq, LDT);
-      tile_write_t<HD, NV>(lds_dot + nxt * HD * LDT, rdo, LDT);
+      tile_write_t<HD, NV>(lds_qt + nxt * HD * TILE, rq);
+      tile_write_t<HD, NV>(lds_dot + nxt * HD * TILE, rdo);
       if (qt + 2 < n_q_tiles) {
         tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
         tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
@@ -648,8 +667,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 grid((S + 127) / 128, B * nh), block(512);
   DISPATCH_HD(HDP, {
-    const int LDK = HD + LPAD, LDV = TILE + LPAD, LDP = TILE + LPAD;
-    size_t smem = (2 * TILE * LDK + 2 * HD * LDV + 8 * 16 * LDP) * sizeof(__bf16);
+    const int LDK = HD + LPAD, LDP = TILE + LPAD;
+    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + 8 * 16 * LDP) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
@@ -679,7 +698,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
   dim3 block(512);
   DISPATCH_HD(HDP, {
     const int LDK = HD + LPAD, LDT = TILE + LPAD;
-    size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * LDT + 8 * 16 * LDT) * sizeof(__bf16);
+    size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dq, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
@@ -689,7 +708,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
     HIP_CHECK_LAST();
     constexpr int NBUF = (HD <= 64) ? 2 : 1;
     size_t smem_dkdv =
-        (NBUF * TILE * LDK * 2 + NBUF * HD * LDT * 2 + 8 * 16 * LDT) * sizeof(__bf16);
+        (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 8 * 16 * LDT) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dkdv, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
