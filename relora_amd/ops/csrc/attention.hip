@@ -211,6 +211,19 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     // diagonal or the sequence end
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
+    // stage tile kt+1 into the other buffer DURING compute: its last
+    // readers finished at the barrier that ended tile kt-1, and no wave
+    // enters tile kt+1 before this tile's end barrier — so one barrier per
+    // tile suffices and the LDS writes overlap the MFMA phase
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+    }
+
     float p_val[4][4];
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
@@ -276,16 +289,6 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
             a, ldsT_frag(vb, t * 16 + col, ks * 32 + kgrp * 8), o_acc[t], 0, 0, 0);
     }
     __syncthreads();
-
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-      __syncthreads();
-    }
   }
 
   // epilogue: O = o_acc / l, LSE = m + log(l)
@@ -399,6 +402,16 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const __bf16* ktb = lds_kt + cur * HD * TILE;
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
+    if (kt + 1 < n_tiles) {  // overlapped staging (see fwd comment)
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+    }
+
     float ds_val[4][4];
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
@@ -443,17 +456,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
             a, ldsT_frag(ktb, t * 16 + col, ks * 32 + kgrp * 8), dq_acc[t], 0, 0, 0);
     }
     __syncthreads();
-
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
-      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-      __syncthreads();
-    }
   }
 
 #pragma unroll
@@ -548,6 +550,17 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const __bf16* dotb = lds_dot + cur * HD * TILE;
     const bool edge = (q_start < kv_start_blk + 127) || (q_start + TILE > S);
 
+    if (NBUF == 2 && qt + 1 < n_q_tiles) {  // overlapped staging (see fwd)
+      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
+      tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+      tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+      if (qt + 2 < n_q_tiles) {
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+      }
+    }
+
     // T = K Q^T (scores transposed), dPT = V dO^T
     float pt_val[4][4], dst_val[4][4];
 #pragma unroll
@@ -610,12 +623,11 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     }
     __syncthreads();
 
-    if (qt + 1 < n_q_tiles) {
-      const int nxt = (NBUF == 2) ? (cur ^ 1) : 0;
-      tile_write_rows<HD, NV>(lds_q + nxt * TILE * LDK, rq, LDK);
-      tile_write_rows<HD, NV>(lds_do + nxt * TILE * LDK, rdo, LDK);
-      tile_write_t<HD, NV>(lds_qt + nxt * HD * TILE, rq);
-      tile_write_t<HD, NV>(lds_dot + nxt * HD * TILE, rdo);
+    if (NBUF == 1 && qt + 1 < n_q_tiles) {
+      tile_write_rows<HD, NV>(lds_q, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do, rdo, LDK);
+      tile_write_t<HD, NV>(lds_qt, rq);
+      tile_write_t<HD, NV>(lds_dot, rdo);
       if (qt + 2 < n_q_tiles) {
         tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
         tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);

@@ -161,25 +161,51 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
     }
   }
 
-  // epilogue: out[m][n] += acc (bf16 read-modify-write), optional mask scale
+  // epilogue in two phases: dump the C-layout accumulators into an LDS
+  // [128][136] tile (cheap strided b16 writes), then vectorized bf16x8
+  // global read-modify-writes — the per-element C-layout RMW was 64 scalar
+  // global round trips per thread
+  __syncthreads();  // done with p_im/q_im; reuse the LDS as the out tile
+  __bf16* o_im = (__bf16*)smem;  // [128][OLD]
+  constexpr int OLD = 128 + LPAD;
   const int crow = (lane >> 4) * 4;  // C-frag rows crow..crow+3, col = fr
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
+    for (int ni = 0; ni < 4; ++ni)
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const long m = m0 + wr + mi * 16 + crow + j;
-        const int n = n0 + wc + ni * 16 + fr;
-        if (m < M && n < N) {
-          float v = acc[mi][ni][j];
-          if (MASK) {
-            const uint8_t mb = mask[m * (long)(N >> 3) + (n >> 3)];
-            v = (mb >> (n & 7)) & 1 ? v * inv_keep : 0.f;
-          }
-          __hip_bfloat16* o = out + m * (long)N + n;
-          *o = from_f32<__hip_bfloat16>(to_f32(*o) + v);
+      for (int j = 0; j < 4; ++j)
+        o_im[(wr + mi * 16 + crow + j) * OLD + wc + ni * 16 + fr] =
+            (__bf16)acc[mi][ni][j];
+  __syncthreads();
+
+  for (int t = threadIdx.x; t < 128 * 16; t += blockDim.x) {
+    const int row = t / 16;
+    const int c8 = (t % 16) * 8;
+    const long m = m0 + row;
+    const int n = n0 + c8;
+    if (m >= M || n >= N) continue;
+    const __bf16* src_v = o_im + row * OLD + c8;
+    __hip_bfloat16* o = out + m * (long)N + n;
+    if (n + 8 <= N) {
+      Vec8<__hip_bfloat16> ov = load8(o);
+      uint8_t mb = 0xFF;
+      if (MASK) mb = mask[m * (long)(N >> 3) + (n >> 3)];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = (float)src_v[j];
+        if (MASK) v = (mb >> j) & 1 ? v * inv_keep : 0.f;
+        ov.v[j] = from_f32<__hip_bfloat16>(to_f32(ov.v[j]) + v);
+      }
+      store8(o, ov);
+    } else {
+      for (int j = 0; j < 8 && n + j < N; ++j) {
+        float v = (float)src_v[j];
+        if (MASK) {
+          const uint8_t mb = mask[m * (long)(N >> 3) + ((n + j) >> 3)];
+          v = (mb >> ((n + j) & 7)) & 1 ? v * inv_keep : 0.f;
         }
+        o[j] = from_f32<__hip_bfloat16>(to_f32(o[j]) + v);
       }
     }
   }
@@ -216,7 +242,8 @@ void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q) {
   TORCH_CHECK(P.size(0) == M && Q.size(0) == N && Q.size(1) == r);
   TORCH_CHECK(r % 32 == 0 && r <= 256, "lora_add: r must be a multiple of 32, <=256");
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  const size_t lds = 2u * 128 * (r + LPAD) * sizeof(__bf16);
+  const size_t lds = std::max<size_t>(2u * 128 * (r + LPAD), 128u * (128 + LPAD))
+                     * sizeof(__bf16);
   dim3 grid((N + 127) / 128, (M + 127) / 128), block(256);
   hipLaunchKernelGGL((lora_skinny_kernel<false, false>), grid, block, lds, stream,
                      (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)Q.data_ptr(),
@@ -235,7 +262,8 @@ void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
   TORCH_CHECK(P.size(0) == M && Q.size(0) == r && Q.size(1) == N);
   TORCH_CHECK(r % 32 == 0 && r <= 256, "lora_add: r must be a multiple of 32, <=256");
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  const size_t lds = 2u * 128 * (r + LPAD) * sizeof(__bf16);
+  const size_t lds = std::max<size_t>(2u * 128 * (r + LPAD), 128u * (128 + LPAD))
+                     * sizeof(__bf16);
   dim3 grid((N + 127) / 128, (M + 127) / 128), block(256);
   const bool has_mask = mask.defined() && mask.numel() > 0;
   if (has_mask) {

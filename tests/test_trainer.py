@@ -156,3 +156,71 @@ def test_args_yaml_override(tmp_path):
     assert args.batch_size == 2
     assert args.lr == 5e-4
     assert isinstance(args.lr, float)
+
+
+# ---------------------------------------------------------------------------
+# the reference README.dev.md regime matrix as scripted smoke tests
+# (SURVEY.md §4 item 5): each regime must complete a short run on CPU
+# ---------------------------------------------------------------------------
+
+
+def _pythia_config(tmp_path):
+    import json as _json
+    cfg = {
+        "architectures": ["GPTNeoXForCausalLM"], "model_type": "gpt_neox",
+        "hidden_size": 32, "intermediate_size": 128, "num_attention_heads": 4,
+        "num_hidden_layers": 2, "vocab_size": 128, "max_position_embeddings": 64,
+        "rotary_pct": 0.25, "rotary_emb_base": 10000, "use_parallel_residual": True,
+        "layer_norm_eps": 1e-5, "initializer_range": 0.02, "tie_word_embeddings": False,
+    }
+    p = tmp_path / "pythia_tiny.json"
+    p.write_text(_json.dumps(cfg))
+    return str(p)
+
+
+def test_regime_pythia_zero_relora(tmp_path):
+    args = run_args(tmp_path, extra=[
+        "--model_config", _pythia_config(tmp_path),
+        "--optimizer", "adam_zero",
+    ], steps=4)
+    main(args)
+    assert (tmp_path / "run" / "model_4" / "training_state.json").exists()
+
+
+def test_regime_pythia_full_rank(tmp_path):
+    args = parse_args([
+        "--model_config", _pythia_config(tmp_path),
+        "--synthetic_data", "true",
+        "--num_training_steps", "3",
+        "--batch_size", "2", "--total_batch_size", "2",
+        "--max_length", "32", "--lr", "1e-3", "--dtype", "float32",
+        "--eval_every", "100", "--save_every", "100", "--workers", "0",
+        "--save_dir", str(tmp_path / "run"),
+    ])
+    main(args)
+    assert (tmp_path / "run" / "model_3").exists()
+
+
+def test_regime_relora_magnitude_pruning_warm_start(tmp_path):
+    # stage 1: short full-rank warmup checkpoint
+    args = parse_args([
+        "--model_config", "configs/llama_9m.json",
+        "--synthetic_data", "true",
+        "--num_training_steps", "2",
+        "--batch_size", "2", "--total_batch_size", "2",
+        "--max_length", "32", "--lr", "1e-3", "--dtype", "float32",
+        "--eval_every", "100", "--save_every", "2", "--workers", "0",
+        "--save_dir", str(tmp_path / "warm"),
+    ])
+    main(args)
+    warm = tmp_path / "warm" / "model_2"
+    assert warm.exists()
+    # stage 2: relora from the warm checkpoint with magnitude pruning
+    args = run_args(tmp_path, extra=[
+        "--warmed_up_model", str(warm),
+        "--optimizer_magnitude_pruning", "0.9",
+    ], steps=8)
+    args.reset_optimizer_on_relora = False
+    main(args)
+    state = json.load(open(tmp_path / "run" / "model_8" / "training_state.json"))
+    assert state["n_optimizer_resets"] >= 1
