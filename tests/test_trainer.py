@@ -182,9 +182,9 @@ def test_regime_pythia_zero_relora(tmp_path):
     args = run_args(tmp_path, extra=[
         "--model_config", _pythia_config(tmp_path),
         "--optimizer", "adam_zero",
-    ], steps=4)
+    ], steps=6)
     main(args)
-    assert (tmp_path / "run" / "model_4" / "training_state.json").exists()
+    assert (tmp_path / "run" / "model_6" / "training_state.json").exists()
 
 
 def test_regime_pythia_full_rank(tmp_path):
@@ -218,9 +218,9 @@ def test_regime_relora_magnitude_pruning_warm_start(tmp_path):
     # stage 2: relora from the warm checkpoint with magnitude pruning
     args = run_args(tmp_path, extra=[
         "--warmed_up_model", str(warm),
+        "--reset_optimizer_on_relora", "false",
         "--optimizer_magnitude_pruning", "0.9",
-    ], steps=8)
-    args.reset_optimizer_on_relora = False
+    ], steps=9)
     main(args)
-    state = json.load(open(tmp_path / "run" / "model_8" / "training_state.json"))
+    state = json.load(open(tmp_path / "run" / "model_9" / "training_state.json"))
     assert state["n_optimizer_resets"] >= 1
