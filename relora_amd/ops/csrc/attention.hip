@@ -211,19 +211,6 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     // diagonal or the sequence end
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
-    // stage tile kt+1 into the other buffer DURING compute: its last
-    // readers finished at the barrier that ended tile kt-1, and no wave
-    // enters tile kt+1 before this tile's end barrier — so one barrier per
-    // tile suffices and the LDS writes overlap the MFMA phase
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-    }
-
     float p_val[4][4];
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
@@ -244,6 +231,20 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
       } else {
 #pragma unroll
         for (int reg = 0; reg < 4; ++reg) p_val[n][reg] = acc[reg] * scale;
+      }
+    }
+
+    // stage tile kt+1 into the other buffer DURING compute: its last
+    // readers finished at the barrier that ended tile kt-1, and no wave
+    // enters tile kt+1 before this tile's end barrier — one barrier per
+    // tile, LDS writes overlapped with the softmax/PV phase.  Placed after
+    // the QK^T cluster so the MFMA chain is not delayed behind staging.
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
       }
     }
 
@@ -402,16 +403,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const __bf16* ktb = lds_kt + cur * HD * TILE;
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
-    if (kt + 1 < n_tiles) {  // overlapped staging (see fwd comment)
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
-      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-    }
-
     float ds_val[4][4];
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
@@ -436,6 +427,16 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
           p = __expf(s_acc[reg] * scale - lse_r[reg]);
         }
         ds_val[n][reg] = p * (dp_acc[reg] - delta_r[reg]) * scale;
+      }
+    }
+
+    if (kt + 1 < n_tiles) {  // overlapped staging, after the MFMA cluster
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
       }
     }
 
@@ -550,17 +551,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const __bf16* dotb = lds_dot + cur * HD * TILE;
     const bool edge = (q_start < kv_start_blk + 127) || (q_start + TILE > S);
 
-    if (NBUF == 2 && qt + 1 < n_q_tiles) {  // overlapped staging (see fwd)
-      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
-      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
-      tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
-      tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
-      if (qt + 2 < n_q_tiles) {
-        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
-      }
-    }
-
     // T = K Q^T (scores transposed), dPT = V dO^T
     float pt_val[4][4], dst_val[4][4];
 #pragma unroll
@@ -589,6 +579,17 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
         }
         pt_val[n][reg] = p;
         dst_val[n][reg] = p * (dpt_acc[reg] - delta_c) * scale;
+      }
+    }
+
+    if (NBUF == 2 && qt + 1 < n_q_tiles) {  // overlapped staging, mid-compute
+      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
+      tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+      tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+      if (qt + 2 < n_q_tiles) {
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
       }
     }
 

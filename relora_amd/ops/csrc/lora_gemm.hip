@@ -58,25 +58,37 @@ DEV_INLINE void philox4(uint64_t seed, uint64_t idx, uint32_t out[4]) {
 // One thread per 8 consecutive elements (one mask byte, one bf16x8 store).
 template <typename T>
 __global__ void dropout_mask_kernel(const T* __restrict__ x, T* __restrict__ xd,
-                                    uint8_t* __restrict__ mask, long n8,
+                                    uint8_t* __restrict__ mask, long n8, long n,
                                     uint64_t seed, float p, float inv_keep) {
   const long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i8 >= n8) return;
   uint32_t r[4];
   philox4(seed, (uint64_t)i8, r);
   const uint32_t thr = (uint32_t)(p * 65536.0f);
-  Vec8<T> v = load8(x + i8 * 8);
-  Vec8<T> o;
-  uint8_t m = 0;
+  const uint32_t u16s[8] = {r[0] & 0xFFFFu, r[0] >> 16, r[1] & 0xFFFFu, r[1] >> 16,
+                            r[2] & 0xFFFFu, r[2] >> 16, r[3] & 0xFFFFu, r[3] >> 16};
+  if (i8 * 8 + 8 <= n) {
+    Vec8<T> v = load8(x + i8 * 8);
+    Vec8<T> o;
+    uint8_t m = 0;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const uint32_t u16 = (r[j >> 1] >> ((j & 1) * 16)) & 0xFFFFu;
-    const bool keep = u16 >= thr;
-    m |= (uint8_t)keep << j;
-    o.v[j] = keep ? from_f32<T>(to_f32(v.v[j]) * inv_keep) : from_f32<T>(0.f);
+    for (int j = 0; j < 8; ++j) {
+      const bool keep = u16s[j] >= thr;
+      m |= (uint8_t)keep << j;
+      o.v[j] = keep ? from_f32<T>(to_f32(v.v[j]) * inv_keep) : from_f32<T>(0.f);
+    }
+    store8(xd + i8 * 8, o);
+    mask[i8] = m;
+  } else {  // tail: < 8 elements
+    uint8_t m = 0;
+    for (int j = 0; j < 8 && i8 * 8 + j < n; ++j) {
+      const bool keep = u16s[j] >= thr;
+      m |= (uint8_t)keep << j;
+      xd[i8 * 8 + j] = keep ? from_f32<T>(to_f32(x[i8 * 8 + j]) * inv_keep)
+                            : from_f32<T>(0.f);
+    }
+    mask[i8] = m;
   }
-  store8(xd + i8 * 8, o);
-  mask[i8] = m;
 }
 
 // ---------------------------------------------------------------------------
@@ -186,11 +198,12 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
     const int n = n0 + c8;
     if (m >= M || n >= N) continue;
     const __bf16* src_v = o_im + row * OLD + c8;
-    __hip_bfloat16* o = out + m * (long)N + n;
-    if (n + 8 <= N) {
+    const long flat = m * (long)N + n;  // mask bits are FLAT-packed over [M*N]
+    __hip_bfloat16* o = out + flat;
+    if (n + 8 <= N && (flat & 7) == 0) {
       Vec8<__hip_bfloat16> ov = load8(o);
       uint8_t mb = 0xFF;
-      if (MASK) mb = mask[m * (long)(N >> 3) + (n >> 3)];
+      if (MASK) mb = mask[flat >> 3];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float v = (float)src_v[j];
@@ -198,12 +211,12 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
         ov.v[j] = from_f32<__hip_bfloat16>(to_f32(ov.v[j]) + v);
       }
       store8(o, ov);
-    } else {
+    } else {  // unaligned rows (odd N) or the row tail
       for (int j = 0; j < 8 && n + j < N; ++j) {
         float v = (float)src_v[j];
         if (MASK) {
-          const uint8_t mb = mask[m * (long)(N >> 3) + ((n + j) >> 3)];
-          v = (mb >> ((n + j) & 7)) & 1 ? v * inv_keep : 0.f;
+          const uint8_t mb = mask[(flat + j) >> 3];
+          v = (mb >> ((flat + j) & 7)) & 1 ? v * inv_keep : 0.f;
         }
         o[j] = from_f32<__hip_bfloat16>(to_f32(o[j]) + v);
       }
@@ -217,17 +230,17 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
 
 std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t seed) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
-  TORCH_CHECK(x.numel() % 8 == 0, "dropout_mask_fwd: numel must be a multiple of 8");
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "dropout_mask_fwd: bf16 only");
   auto xd = torch::empty_like(x);
-  const long n8 = x.numel() / 8;
+  const long n = x.numel();
+  const long n8 = (n + 7) / 8;  // FLAT packing: mask bit j of byte b = elem 8b+j
   auto mask = torch::empty({n8}, x.options().dtype(torch::kUInt8));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const float inv_keep = 1.f / (1.f - (float)p);
   hipLaunchKernelGGL(dropout_mask_kernel<__hip_bfloat16>,
                      dim3((n8 + 255) / 256), dim3(256), 0, stream,
                      (const __hip_bfloat16*)x.data_ptr(), (__hip_bfloat16*)xd.data_ptr(),
-                     mask.data_ptr<uint8_t>(), n8, (uint64_t)seed, (float)p, inv_keep);
+                     mask.data_ptr<uint8_t>(), n8, n, (uint64_t)seed, (float)p, inv_keep);
   HIP_CHECK_LAST();
   return {xd, mask};
 }
@@ -267,8 +280,7 @@ void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
   dim3 grid((N + 127) / 128, (M + 127) / 128), block(256);
   const bool has_mask = mask.defined() && mask.numel() > 0;
   if (has_mask) {
-    TORCH_CHECK(N % 8 == 0, "masked lora_add: N must be a multiple of 8");
-    TORCH_CHECK(mask.numel() == M * (long)(N / 8), "mask size mismatch");
+    TORCH_CHECK(mask.numel() == (M * (long)N + 7) / 8, "mask size mismatch");
     hipLaunchKernelGGL((lora_skinny_kernel<true, true>), grid, block, lds, stream,
                        (const __hip_bfloat16*)P.data_ptr(),
                        (const __hip_bfloat16*)Q.data_ptr(),
@@ -336,13 +348,13 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
       const int mm = t / 16;
       const int c8 = (t % 16) * 8;
       bf16x8 v;
-      if (m0 + mm < m_end && c0 + c8 + 8 <= C) {
-        v = *reinterpret_cast<const bf16x8*>(X + (m0 + mm) * (long)C + c0 + c8);
+      const long xoff = (m0 + mm) * (long)C + c0 + c8;
+      if (m0 + mm < m_end && c0 + c8 + 8 <= C && (xoff & 7) == 0) {
+        v = *reinterpret_cast<const bf16x8*>(X + xoff);
       } else if (m0 + mm < m_end) {
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          v[j] = (c0 + c8 + j < C) ? (__bf16)X[(m0 + mm) * (long)C + c0 + c8 + j]
-                                   : (__bf16)0.f;
+          v[j] = (c0 + c8 + j < C) ? (__bf16)X[xoff + j] : (__bf16)0.f;
       } else {
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }

@@ -400,7 +400,6 @@ def _fused_ok(x, weight, lora_A, scale, lora_only):
     return (hip.use_hip(x) and not lora_only and not torch.is_tensor(scale)
             and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
             and r % 32 == 0 and r <= 256
-            and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
             and os.environ.get("RELORA_AMD_LORA_PATH", "fused") != "torch")
 
 

@@ -220,7 +220,7 @@ def test_regime_relora_magnitude_pruning_warm_start(tmp_path):
         "--warmed_up_model", str(warm),
         "--reset_optimizer_on_relora", "false",
         "--optimizer_magnitude_pruning", "0.9",
-    ], steps=9)
+    ], steps=8)
     main(args)
-    state = json.load(open(tmp_path / "run" / "model_9" / "training_state.json"))
+    state = json.load(open(tmp_path / "run" / "model_8" / "training_state.json"))
     assert state["n_optimizer_resets"] >= 1
