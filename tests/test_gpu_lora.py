@@ -223,3 +223,39 @@ def test_quantized_relora_linear_gpu():
     lin.merge_and_reinit()
     post = lin._dense_weight()
     assert not torch.equal(pre, post)
+
+
+def test_fused_lora_linear_odd_dims():
+    """llama_1b's intermediate_size is 5461 (odd) — gate/up (N odd) and
+    down (K odd) must still route through the fused kernels correctly."""
+    torch.manual_seed(8)
+    for (M, K, N) in ((256, 2048, 341), (256, 341, 512)):
+        x = (torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.5).requires_grad_(True)
+        W = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.02
+        A = (torch.randn(64, K, device="cuda", dtype=torch.bfloat16) * 0.02).requires_grad_(True)
+        B = (torch.randn(N, 64, device="cuda", dtype=torch.bfloat16) * 0.02).requires_grad_(True)
+        y = _FusedLoRALinear.apply(x, W, None, A, B, 0.5, 0.0, True)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+
+        x2 = x.detach().clone().requires_grad_(True)
+        A2 = A.detach().clone().requires_grad_(True)
+        B2 = B.detach().clone().requires_grad_(True)
+        os.environ["RELORA_AMD_LORA_PATH"] = "torch"
+        try:
+            y2 = lora_linear(x2, W, None, A2, B2, 0.5, dropout_p=0.0, training=True)
+        finally:
+            del os.environ["RELORA_AMD_LORA_PATH"]
+        y2.backward(dy)
+        for got, ref, name in ((y, y2, "y"), (x.grad, x2.grad, "dx"),
+                               (A.grad, A2.grad, "dA"), (B.grad, B2.grad, "dB")):
+            err = (got.float() - ref.float()).abs().max().item()
+            tol = 0.05 * max(1.0, ref.float().abs().max().item())
+            assert err < tol, f"{M}x{K}x{N} {name}: {err} vs {tol}"
+
+        # with dropout: runs and produces sane sparsity in dx
+        x3 = x.detach().clone().requires_grad_(True)
+        y3 = _FusedLoRALinear.apply(x3, torch.zeros_like(W), None, A, B, 1.0, 0.3, True)
+        y3.backward(torch.randn_like(y3))
+        frac = (x3.grad == 0).float().mean().item()
+        assert abs(frac - 0.3) < 0.06, frac
