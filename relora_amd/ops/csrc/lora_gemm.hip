@@ -400,8 +400,30 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
   }
 }
 
-// out[r, C] fp32 = P[M,r]^T @ X[M,C] as MCHUNK partial planes summed by torch
-torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X) {
+// combine the MCHUNK fp32 partial planes: out = scale * sum_chunks(part),
+// emitted in the requested dtype, optionally transposed ([C, r] instead of
+// [r, C]) — folds the .sum(0).t().contiguous().mul_(scale).to(bf16) chain
+// (5 launches per wrapped Linear) into one kernel.
+template <typename T, bool TRANSP>
+__global__ void skinny_combine_kernel(const float* __restrict__ part,
+                                      T* __restrict__ out, int r, int C,
+                                      int nchunks, float scale) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= (long)r * C) return;
+  float acc = 0.f;
+  for (int c = 0; c < nchunks; ++c) acc += part[(long)c * r * C + i];
+  acc *= scale;
+  if (TRANSP) {
+    const int rr = i / C, cc = i % C;
+    out[(long)cc * r + rr] = from_f32<T>(acc);
+  } else {
+    out[i] = from_f32<T>(acc);
+  }
+}
+
+// out = scale * (P[M,r]^T @ X[M,C]) in `dtype`; [C, r] when transpose_out
+torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
+                          bool transpose_out, torch::ScalarType dtype) {
   TORCH_CHECK(P.is_cuda() && P.is_contiguous() && X.is_contiguous());
   TORCH_CHECK(P.scalar_type() == torch::kBFloat16 && X.scalar_type() == torch::kBFloat16);
   const long M = P.size(0);
@@ -425,5 +447,29 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X) {
                      (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
                      part.data_ptr<float>(), M, C, r, rows);
   HIP_CHECK_LAST();
-  return part.sum(0);
+  auto out = transpose_out
+      ? torch::empty({C, r}, P.options().dtype(dtype))
+      : torch::empty({r, C}, P.options().dtype(dtype));
+  dim3 cgrid(((long)r * C + 255) / 256), cblock(256);
+  const float s = (float)scale;
+  if (dtype == torch::kBFloat16) {
+    if (transpose_out)
+      hipLaunchKernelGGL((skinny_combine_kernel<__hip_bfloat16, true>), cgrid, cblock, 0,
+                         stream, part.data_ptr<float>(),
+                         (__hip_bfloat16*)out.data_ptr(), r, C, chunks, s);
+    else
+      hipLaunchKernelGGL((skinny_combine_kernel<__hip_bfloat16, false>), cgrid, cblock, 0,
+                         stream, part.data_ptr<float>(),
+                         (__hip_bfloat16*)out.data_ptr(), r, C, chunks, s);
+  } else {
+    TORCH_CHECK(dtype == torch::kFloat32, "skinny_grad: bf16 or fp32 output");
+    if (transpose_out)
+      hipLaunchKernelGGL((skinny_combine_kernel<float, true>), cgrid, cblock, 0, stream,
+                         part.data_ptr<float>(), out.data_ptr<float>(), r, C, chunks, s);
+    else
+      hipLaunchKernelGGL((skinny_combine_kernel<float, false>), cgrid, cblock, 0, stream,
+                         part.data_ptr<float>(), out.data_ptr<float>(), r, C, chunks, s);
+  }
+  HIP_CHECK_LAST();
+  return out;
 }

@@ -386,9 +386,10 @@ class _FusedLoRALinear(torch.autograd.Function):
                                1.0 / (1.0 - p) if p > 0 else 1.0)
 
         # dA = u_s^T @ xd, dB = s * dy^T @ t_u = (t_u^T @ dy)^T * s — via the
-        # chunked-M skinny_grad kernel (hipBLASLt launches these 32-WG wide)
-        dA = hip.ext().skinny_grad(u_s, xd).to(lora_A.dtype)
-        dB = hip.ext().skinny_grad(t_u, dy2d).t().contiguous().mul_(scale).to(bs.dtype)
+        # chunked-M skinny_grad kernel (hipBLASLt launches these 32-WG wide);
+        # scale/transpose/cast fold into its combine stage
+        dA = hip.ext().skinny_grad(u_s, xd, 1.0, False, lora_A.dtype)
+        dB = hip.ext().skinny_grad(t_u, dy2d, scale, True, bs.dtype)
         dw = dy2d.t() @ x2d if ctx.needs_input_grad[1] else None
         dbias = dy2d.sum(0) if ctx.has_bias and ctx.needs_input_grad[2] else None
         return (dx.view(dy.shape[:-1] + (weight.shape[1],)), dw, dbias,
@@ -397,9 +398,13 @@ class _FusedLoRALinear(torch.autograd.Function):
 
 def _fused_ok(x, weight, lora_A, scale, lora_only):
     r = lora_A.shape[0]
+    # odd in/out dims (llama_1b intermediate 5461) are CORRECT through the
+    # fused kernels (alignment-guarded fallbacks) but measured slower than
+    # the hipBLASLt composition — keep them on the library path
     return (hip.use_hip(x) and not lora_only and not torch.is_tensor(scale)
             and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
             and r % 32 == 0 and r <= 256
+            and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
             and os.environ.get("RELORA_AMD_LORA_PATH", "fused") != "torch")
 
 

@@ -166,12 +166,17 @@ def test_skinny_grad_matches_matmul():
     for M, r, C in ((4096, 128, 2048), (16384, 128, 512), (1000, 64, 320)):
         P = torch.randn(M, r, device="cuda", dtype=torch.bfloat16) * 0.1
         X = torch.randn(M, C, device="cuda", dtype=torch.bfloat16) * 0.1
-        got = ext().skinny_grad(P, X)
+        got = ext().skinny_grad(P, X, 1.0, False, torch.float32)
         ref = P.float().t() @ X.float()
         # bf16 products, fp32 accumulation: tolerance scales with sqrt(M)
         tol = 0.03 * ref.abs().max().item() + 0.05
         err = (got - ref).abs().max().item()
         assert err < tol, (M, r, C, err, tol)
+        # transposed + scaled + bf16 variant
+        gt = ext().skinny_grad(P, X, 0.5, True, torch.bfloat16)
+        assert gt.shape == (C, r)
+        errt = (gt.float() - 0.5 * ref.t()).abs().max().item()
+        assert errt < tol, (M, r, C, errt)
 
 
 def test_quantize_kernels_match_refs():
