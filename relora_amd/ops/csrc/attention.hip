@@ -163,7 +163,9 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   __bf16* lds_p = lds_vt + 2 * HD * TILE;     // [8][16][LDP]
 
   const int bh = blockIdx.y;
-  const int q_start = blockIdx.x * 128;
+  // heavy blocks (high q_start: up to 16x the kv tiles of block 0) first,
+  // so the causal work imbalance doesn't leave a long tail
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * 128;
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -348,7 +350,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   __bf16* lds_p = lds_kt + 2 * HD * TILE;     // [8][16][LDT]
 
   const int bh = blockIdx.y;
-  const int q_start = blockIdx.x * 128;
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * 128;  // heavy blocks first
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
