@@ -498,7 +498,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   __bf16* lds_do = lds_q + NBUF * TILE * LDK;     // [NBUF][TILE][LDK]
   __bf16* lds_qt = lds_do + NBUF * TILE * LDK;    // [NBUF][HD][TILE] rotated
   __bf16* lds_dot = lds_qt + NBUF * HD * TILE;    // [NBUF][HD][TILE] rotated
-  __bf16* lds_p = lds_dot + NBUF * HD * TILE;     // [8][16][LDT]
+  __bf16* lds_p = lds_dot + NBUF * HD * TILE;     // [8][16][LDT]  P^T
+  __bf16* lds_p2 = lds_p + 8 * 16 * (TILE + LPAD);  // [8][16][LDT]  dS^T
 
   const int bh = blockIdx.y;
   const int kv_start_blk = blockIdx.x * 128;
@@ -595,34 +596,29 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       }
     }
 
+    // write BOTH relayouts up front into separate per-wave regions, then run
+    // both MFMA groups back-to-back: one lgkm boundary and no
+    // write-after-read stall between the dV and dK passes
     __bf16* pw = lds_p + wave * 16 * LDT;
-    // pass 1: dV += P^T @ dO  (intra-wave relayout, no block barrier)
+    __bf16* pw2 = lds_p2 + wave * 16 * LDT;
 #pragma unroll
     for (int n = 0; n < 4; ++n)
 #pragma unroll
-      for (int reg = 0; reg < 4; ++reg)
+      for (int reg = 0; reg < 4; ++reg) {
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)pt_val[n][reg];
+        pw2[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)dst_val[n][reg];
+      }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
+      const bf16x8 a2 = lds_frag(pw2, col, ks * 32 + kgrp * 8, LDT);
 #pragma unroll
-      for (int t = 0; t < NT_HD; ++t)
+      for (int t = 0; t < NT_HD; ++t) {
         dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, ldsT_frag(dotb, t * 16 + col, ks * 32 + kgrp * 8), dv_acc[t], 0, 0, 0);
-    }
-    // pass 2: dK += dS^T @ Q
-#pragma unroll
-    for (int n = 0; n < 4; ++n)
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg)
-        pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)dst_val[n][reg];
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
-#pragma unroll
-      for (int t = 0; t < NT_HD; ++t)
         dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, ldsT_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8), dk_acc[t], 0, 0, 0);
+            a2, ldsT_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8), dk_acc[t], 0, 0, 0);
+      }
     }
     __syncthreads();
 
@@ -723,7 +719,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
     HIP_CHECK_LAST();
     constexpr int NBUF = (HD <= 64) ? 2 : 1;
     size_t smem_dkdv =
-        (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 8 * 16 * LDT) * sizeof(__bf16);
+        (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 2 * 8 * 16 * LDT) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dkdv, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
