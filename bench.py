@@ -59,11 +59,14 @@ def main():
         dist.init_process_group(backend=backend, rank=rank, world_size=world_size)
 
     from relora_amd.models import build_model_from_config, load_model_config
+    from relora_amd.ops.tunable import enable_tuned_gemms
     from relora_amd.ops.optim import AdamW, clip_grad_norm_
     from relora_amd.parallel import DistributedModel
     from relora_amd.relora import ReLoRaModel
 
     torch.manual_seed(1234)
+    if use_gpu:
+        enable_tuned_gemms()
     cfg = load_model_config(args.model)
     model = build_model_from_config(cfg)
     if not args.full_rank:

@@ -37,6 +37,7 @@ from relora_amd.models import build_model_from_config, load_model_config
 from relora_amd.models.llama import LlamaForCausalLM
 from relora_amd.models.pythia import GPTNeoXForCausalLM
 from relora_amd.ops.optim import AdamW, clip_grad_norm_
+from relora_amd.ops.tunable import enable_tuned_gemms
 from relora_amd.parallel import DistributedModel, ZeroRedundancyAdamW
 from relora_amd.relora import ReLoRaLinear, ReLoRaModel
 from relora_amd.utils.logging import logger
@@ -256,6 +257,7 @@ def main(args):
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         torch.cuda.set_device(local_rank)
+        enable_tuned_gemms()
         device = f"cuda:{local_rank}"
         backend = "nccl"  # = RCCL on ROCm
     else:
