@@ -158,10 +158,8 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   constexpr int NV = (HD + 63) / 64;   // bf16x8 staging slices per thread
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // K is NOT staged: QK^T B-fragments read straight from global — the
-  // 16 KB tile is L1/L2-resident across the block's 8 waves, and dropping
-  // the staging doubles residency (2 -> 4 blocks/CU of barrier overlap)
-  __bf16* lds_vt = (__bf16*)smem;             // [2][HD][TILE] rotated
+  __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK]
+  __bf16* lds_vt = lds_k + 2 * TILE * LDK;    // [2][HD][TILE] rotated
   __bf16* lds_p = lds_vt + 2 * HD * TILE;     // [8][16][LDP]
 
   const int bh = blockIdx.y;
@@ -195,15 +193,21 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   const int q_max_abs = min(q_start + 127, S - 1);
   const int n_tiles = (q_max_abs / TILE) + 1;  // causal bound
 
-  bf16x8 rv[NV];
+  bf16x8 rk[NV], rv[NV];
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
   tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_t<HD, NV>(lds_vt, rv);
-  if (n_tiles > 1) tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
   __syncthreads();
 
   for (int kt = 0; kt < n_tiles; ++kt) {
     const int cur = kt & 1;
     const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
     const __bf16* vb = lds_vt + cur * HD * TILE;
     // per-element causal/valid checks only where the tile crosses the
     // diagonal or the sequence end
@@ -216,8 +220,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
 #pragma unroll
       for (int kf = 0; kf < KFRAGS; ++kf)
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            qfrag[kf], global_frag(kp, kv0 + n * 16 + col, S, hd, kf * 32 + kgrp * 8),
-            acc, 0, 0, 0);
+            qfrag[kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK), acc, 0, 0, 0);
       if (edge) {
         const int kv_abs = kv0 + n * 16 + col;
 #pragma unroll
@@ -239,8 +242,12 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     // tile, LDS writes overlapped with the softmax/PV phase.  Placed after
     // the QK^T cluster so the MFMA chain is not delayed behind staging.
     if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
       tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
-      if (kt + 2 < n_tiles) tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
     }
 
     // online softmax per row (4 regs per lane)
@@ -337,9 +344,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   constexpr int NV = (HD + 63) / 64;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // K/V row tiles are not staged (global B-fragments, L1/L2-resident);
-  // only K^T needs LDS
-  __bf16* lds_kt = (__bf16*)smem;             // [2][HD][TILE]  K transposed (rotated)
+  __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK] K rows
+  __bf16* lds_v = lds_k + 2 * TILE * LDK;     // [2][TILE][LDK] V rows
+  __bf16* lds_kt = lds_v + 2 * TILE * LDK;    // [2][HD][TILE]  K transposed (rotated)
   __bf16* lds_p = lds_kt + 2 * HD * TILE;     // [8][16][LDT]
 
   const int bh = blockIdx.y;
@@ -378,15 +385,23 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int q_max_abs = min(q_start + 127, S - 1);
   const int n_tiles = (q_max_abs / TILE) + 1;
 
-  bf16x8 rk[NV];
+  bf16x8 rk[NV], rv[NV];
   tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
+  tile_write_rows<HD, NV>(lds_v, rv, LDK);
   tile_write_t<HD, NV>(lds_kt, rk);
-  if (n_tiles > 1) tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
   __syncthreads();
 
   for (int kt = 0; kt < n_tiles; ++kt) {
     const int cur = kt & 1;
     const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
+    const __bf16* vb = lds_v + cur * TILE * LDK;
     const __bf16* ktb = lds_kt + cur * HD * TILE;
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
@@ -398,11 +413,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int kf = 0; kf < KFRAGS; ++kf) {
         s_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            qfrag[kf], global_frag(kp, kv0 + n * 16 + col, S, hd, kf * 32 + kgrp * 8),
-            s_acc, 0, 0, 0);
+            qfrag[kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK), s_acc, 0, 0, 0);
         dp_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            dofrag[kf], global_frag(vp, kv0 + n * 16 + col, S, hd, kf * 32 + kgrp * 8),
-            dp_acc, 0, 0, 0);
+            dofrag[kf], lds_frag(vb, n * 16 + col, kf * 32 + kgrp * 8, LDK), dp_acc, 0, 0, 0);
       }
       const int kv_abs = kv0 + n * 16 + col;
 #pragma unroll
@@ -420,8 +433,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     }
 
     if (kt + 1 < n_tiles) {  // overlapped staging, after the MFMA cluster
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
       tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
-      if (kt + 2 < n_tiles) tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
     }
 
     // redistribute dS -> A layout (intra-wave)
@@ -472,11 +490,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   constexpr int LDK = HD + LPAD;
   constexpr int LDT = TILE + LPAD;
   constexpr int NV = (HD + 63) / 64;
-  constexpr int NBUF = 2;  // fits for hd<=128 now that row tiles are unstaged
+  constexpr int NBUF = (HD <= 64) ? 2 : 1;  // hd128: 4 double-buffered tiles
+                                            // exceed 160 KiB LDS
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // Q/dO row tiles read as global B-fragments; Q^T / dO^T stay in LDS
-  __bf16* lds_qt = (__bf16*)smem;                 // [NBUF][HD][TILE] rotated
+  __bf16* lds_q = (__bf16*)smem;                  // [NBUF][TILE][LDK]
+  __bf16* lds_do = lds_q + NBUF * TILE * LDK;     // [NBUF][TILE][LDK]
+  __bf16* lds_qt = lds_do + NBUF * TILE * LDK;    // [NBUF][HD][TILE] rotated
   __bf16* lds_dot = lds_qt + NBUF * HD * TILE;    // [NBUF][HD][TILE] rotated
   __bf16* lds_p = lds_dot + NBUF * HD * TILE;     // [8][16][LDT]  P^T
   __bf16* lds_p2 = lds_p + 8 * 16 * (TILE + LPAD);  // [8][16][LDT]  dS^T
@@ -515,6 +535,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   bf16x8 rq[NV], rdo[NV];
   tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd);
   tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
+  tile_write_rows<HD, NV>(lds_q, rq, LDK);
+  tile_write_rows<HD, NV>(lds_do, rdo, LDK);
   tile_write_t<HD, NV>(lds_qt, rq);
   tile_write_t<HD, NV>(lds_dot, rdo);
   if (first_qt + 1 < n_q_tiles) {
@@ -526,6 +548,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
     const int cur = (NBUF == 2) ? (qt & 1) : 0;
     const int q_start = qt * TILE;
+    const __bf16* qb = lds_q + cur * TILE * LDK;
+    const __bf16* dob = lds_do + cur * TILE * LDK;
     const __bf16* qtb = lds_qt + cur * HD * TILE;
     const __bf16* dotb = lds_dot + cur * HD * TILE;
     const bool edge = (q_start < kv_start_blk + 127) || (q_start + TILE > S);
@@ -539,11 +563,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 #pragma unroll
       for (int kf = 0; kf < KFRAGS; ++kf) {
         t_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            kfrag[kf], global_frag(qp, q_start + n * 16 + col, S, hd, kf * 32 + kgrp * 8),
-            t_acc, 0, 0, 0);
+            kfrag[kf], lds_frag(qb, n * 16 + col, kf * 32 + kgrp * 8, LDK), t_acc, 0, 0, 0);
         dpt_acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            vfrag[kf], global_frag(dop, q_start + n * 16 + col, S, hd, kf * 32 + kgrp * 8),
-            dpt_acc, 0, 0, 0);
+            vfrag[kf], lds_frag(dob, n * 16 + col, kf * 32 + kgrp * 8, LDK), dpt_acc, 0, 0, 0);
       }
       const int q_abs = q_start + n * 16 + col;
       const float lse_c = (q_abs < S) ? lse[(long)bh * S + q_abs] : 0.f;
@@ -564,6 +586,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     }
 
     if (NBUF == 2 && qt + 1 < n_q_tiles) {  // overlapped staging, mid-compute
+      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
       tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
       tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
       if (qt + 2 < n_q_tiles) {
@@ -599,6 +623,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     __syncthreads();
 
     if (NBUF == 1 && qt + 1 < n_q_tiles) {
+      tile_write_rows<HD, NV>(lds_q, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do, rdo, LDK);
       tile_write_t<HD, NV>(lds_qt, rq);
       tile_write_t<HD, NV>(lds_dot, rdo);
       if (qt + 2 < n_q_tiles) {
@@ -652,8 +678,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 grid((S + 127) / 128, B * nh), block(512);
   DISPATCH_HD(HDP, {
-    const int LDP = TILE + LPAD;
-    size_t smem = (2 * HD * TILE + 8 * 16 * LDP) * sizeof(__bf16);
+    const int LDK = HD + LPAD, LDP = TILE + LPAD;
+    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + 8 * 16 * LDP) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
@@ -682,8 +708,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
 
   dim3 block(512);
   DISPATCH_HD(HDP, {
-    const int LDT = TILE + LPAD;
-    size_t smem_dq = (2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
+    const int LDK = HD + LPAD, LDT = TILE + LPAD;
+    size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dq, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
@@ -691,9 +717,9 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                        lse.data_ptr<float>(), delta.data_ptr<float>(),
                        (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
     HIP_CHECK_LAST();
-    constexpr int NBUF = 2;
+    constexpr int NBUF = (HD <= 64) ? 2 : 1;
     size_t smem_dkdv =
-        (NBUF * HD * TILE * 2 + 2 * 8 * 16 * LDT) * sizeof(__bf16);
+        (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 2 * 8 * 16 * LDT) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
                        smem_dkdv, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
