@@ -49,6 +49,8 @@ def check_args_torchrun_main(args, n_cli_args=None):
 
     if args.batch_size is None:
         raise ValueError("batch_size must be specified")
+    if args.batch_size == "auto" and args.total_batch_size is None:
+        raise ValueError("--batch_size auto requires --total_batch_size")
 
     if args.tags is not None and isinstance(args.tags, str):
         args.tags = args.tags.split(",")
@@ -65,11 +67,12 @@ def check_args_torchrun_main(args, n_cli_args=None):
         args.lora_r = None
         args.force_keep_original = False
 
-    if args.total_batch_size is None:
-        args.gradient_accumulation = args.gradient_accumulation or 1
-        args.total_batch_size = args.batch_size * args.gradient_accumulation
-
-    assert args.total_batch_size % args.batch_size == 0, "total_batch_size must be divisible by batch_size"
+    if args.batch_size != "auto":
+        if args.total_batch_size is None:
+            args.gradient_accumulation = args.gradient_accumulation or 1
+            args.total_batch_size = args.batch_size * args.gradient_accumulation
+        assert args.total_batch_size % args.batch_size == 0, \
+            "total_batch_size must be divisible by batch_size"
 
     if args.max_train_tokens is not None:
         args.num_training_steps = args.max_train_tokens // args.total_batch_size

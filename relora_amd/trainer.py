@@ -60,6 +60,10 @@ def default_data_collator(features):
     return batch
 
 
+def _int_or_auto(v):
+    return v if v == "auto" else int(v)
+
+
 def parse_args(args=None):
     parser = argparse.ArgumentParser()
 
@@ -84,7 +88,8 @@ def parse_args(args=None):
                         help="Train on deterministic random tokens (offline benchmarking)")
     parser.add_argument("--max_length", type=int, default=512)
 
-    parser.add_argument("--batch_size", type=int, default=None)
+    parser.add_argument("--batch_size", type=_int_or_auto, default=None,
+                        help="micro-batch per GPU, or 'auto' to size for the device HBM (288 GB on MI355X)")
     parser.add_argument("--gradient_accumulation", type=int, default=None)
     parser.add_argument("--total_batch_size", type=int, default=None)
 
@@ -270,6 +275,19 @@ def main(args):
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29511")
         dist.init_process_group(backend=backend, rank=global_rank, world_size=world_size)  # C1
+
+    if args.batch_size == "auto":
+        # size the micro-batch for this device's HBM (288 GB on MI355X)
+        from relora_amd.utils.memory import auto_micro_batch
+
+        if args.model_config is None:
+            raise ValueError("--batch_size auto requires --model_config")
+        _cfg = load_model_config(args.model_config)
+        args.batch_size = auto_micro_batch(
+            _cfg, args.max_length, args.total_batch_size, world_size,
+            lora_r=(args.lora_r or 0) if args.use_peft else 0,
+            trainable_ratio=0.08 if args.use_peft else 1.0)
+        args.gradient_accumulation = None
 
     if args.total_batch_size is not None and args.gradient_accumulation is None:
         assert args.total_batch_size % world_size == 0, "total_batch_size must be divisible by world_size"

@@ -224,3 +224,37 @@ def test_regime_relora_magnitude_pruning_warm_start(tmp_path):
     main(args)
     state = json.load(open(tmp_path / "run" / "model_8" / "training_state.json"))
     assert state["n_optimizer_resets"] >= 1
+
+
+def test_auto_batch_size(tmp_path):
+    """--batch_size auto sizes the micro-batch for the device and trains."""
+    args = parse_args([
+        "--model_config", "configs/llama_9m.json",
+        "--synthetic_data", "true",
+        "--batch_size", "auto",
+        "--total_batch_size", "8",
+        "--num_training_steps", "2",
+        "--max_length", "32", "--lr", "1e-3", "--dtype", "float32",
+        "--eval_every", "100", "--save_every", "100", "--workers", "0",
+        "--save_dir", str(tmp_path / "run"),
+    ])
+    assert args.batch_size == "auto"
+    main(args)
+    assert (tmp_path / "run" / "model_2").exists()
+
+
+def test_auto_micro_batch_estimator():
+    from relora_amd.models import load_model_config
+    from relora_amd.utils.memory import auto_micro_batch, estimate_step_bytes
+
+    cfg = load_model_config("configs/llama_1b.json")
+    # llama-1b at seq 2048 should fit micro-batch >= 8 in 288 GB
+    bs = auto_micro_batch(cfg, 2048, 1024, 8, lora_r=128,
+                          trainable_ratio=0.08, hbm_bytes=288 << 30)
+    assert bs >= 8, bs
+    # monotone in micro-batch
+    a = estimate_step_bytes(cfg, 4, 2048, lora_r=128)
+    b = estimate_step_bytes(cfg, 8, 2048, lora_r=128)
+    assert b > a
+    # a tiny budget degrades gracefully to 1
+    assert auto_micro_batch(cfg, 2048, 1024, 8, hbm_bytes=1 << 30) == 1
