@@ -214,6 +214,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
     float p_val[4][4];
+    __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -235,6 +236,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
         for (int reg = 0; reg < 4; ++reg) p_val[n][reg] = acc[reg] * scale;
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // stage tile kt+1 into the other buffer DURING compute: its last
     // readers finished at the barrier that ended tile kt-1, and no wave
@@ -283,6 +285,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
         pw[(kgrp * 4 + reg) * LDP + n * 16 + col] = (__bf16)p_val[n][reg];
 
     // PV: two K=32 steps over the 64-row kv tile
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDP);
@@ -291,6 +294,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
         o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, ldsT_frag(vb, t * 16 + col, ks * 32 + kgrp * 8), o_acc[t], 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -406,6 +410,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
     float ds_val[4][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       f32x4 s_acc = {0.f, 0.f, 0.f, 0.f};
@@ -431,6 +436,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
         ds_val[n][reg] = p * (dp_acc[reg] - delta_r[reg]) * scale;
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (kt + 1 < n_tiles) {  // overlapped staging, after the MFMA cluster
       tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
@@ -450,6 +456,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       for (int reg = 0; reg < 4; ++reg)
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)ds_val[n][reg];
 
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
@@ -458,6 +465,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
         dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, ldsT_frag(ktb, t * 16 + col, ks * 32 + kgrp * 8), dq_acc[t], 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -556,6 +564,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 
     // T = K Q^T (scores transposed), dPT = V dO^T
     float pt_val[4][4], dst_val[4][4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
       f32x4 t_acc = {0.f, 0.f, 0.f, 0.f};
@@ -584,6 +593,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
         dst_val[n][reg] = p * (dpt_acc[reg] - delta_c) * scale;
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (NBUF == 2 && qt + 1 < n_q_tiles) {  // overlapped staging, mid-compute
       tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
@@ -608,6 +618,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
         pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)pt_val[n][reg];
         pw2[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)dst_val[n][reg];
       }
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDT);
@@ -620,6 +631,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
             a2, ldsT_frag(qtb, t * 16 + col, ks * 32 + kgrp * 8), dk_acc[t], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
 
     if (NBUF == 1 && qt + 1 < n_q_tiles) {
