@@ -156,16 +156,19 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   constexpr int LDK = HD + LPAD;
   constexpr int LDP = TILE + LPAD;
   constexpr int NV = (HD + 63) / 64;   // bf16x8 staging slices per thread
+  constexpr int RF = 2;                // q row-fragments per wave: 256-row
+                                       // blocks halve K/V traffic + staging
+                                       // per MFMA vs 128-row blocks
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK]
   __bf16* lds_vt = lds_k + 2 * TILE * LDK;    // [2][HD][TILE] rotated
-  __bf16* lds_p = lds_vt + 2 * HD * TILE;     // [8][16][LDP]
+  __bf16* lds_p = lds_vt + 2 * HD * TILE;     // [RF][8][16][LDP]
 
   const int bh = blockIdx.y;
-  // heavy blocks (high q_start: up to 16x the kv tiles of block 0) first,
+  // heavy blocks (high q_start: up to 8x the kv tiles of block 0) first,
   // so the causal work imbalance doesn't leave a long tail
-  const int q_start = (gridDim.x - 1 - blockIdx.x) * 128;
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * (128 * RF);
   const long base = (long)bh * S * hd;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -176,21 +179,31 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   const int col = lane & 15;        // C-frag column / B-frag n / A-frag row
   const int kgrp = lane >> 4;       // k-element group (x8)
 
-  // Q fragments for this wave's 16 rows (A-layout)
-  const int qrow_abs = q_start + wave * 16 + col;
-  bf16x8 qfrag[KFRAGS];
+  // Q fragments: RF row groups of 16 per wave (A-layout); group rf covers
+  // rows q_start + rf*128 + wave*16 + [0,16)
+  bf16x8 qfrag[RF][KFRAGS];
 #pragma unroll
-  for (int kf = 0; kf < KFRAGS; ++kf)
-    qfrag[kf] = global_frag(qp, qrow_abs, S, hd, kf * 32 + kgrp * 8);
-
-  // online softmax state: 4 rows per lane (rows kgrp*4 + reg of the wave tile)
-  float m_run[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
-  float l_run[4] = {0.f, 0.f, 0.f, 0.f};
-  f32x4 o_acc[NT_HD];
+  for (int rf = 0; rf < RF; ++rf)
 #pragma unroll
-  for (int t = 0; t < NT_HD; ++t) o_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int kf = 0; kf < KFRAGS; ++kf)
+      qfrag[rf][kf] = global_frag(qp, q_start + rf * 128 + wave * 16 + col,
+                                  S, hd, kf * 32 + kgrp * 8);
 
-  const int q_max_abs = min(q_start + 127, S - 1);
+  float m_run[RF][4], l_run[RF][4];
+  f32x4 o_acc[RF][NT_HD];
+#pragma unroll
+  for (int rf = 0; rf < RF; ++rf)
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      m_run[rf][reg] = -INFINITY;
+      l_run[rf][reg] = 0.f;
+    }
+#pragma unroll
+  for (int rf = 0; rf < RF; ++rf)
+#pragma unroll
+    for (int t = 0; t < NT_HD; ++t) o_acc[rf][t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int q_max_abs = min(q_start + 128 * RF - 1, S - 1);
   const int n_tiles = (q_max_abs / TILE) + 1;  // causal bound
 
   bf16x8 rk[NV], rv[NV];
@@ -213,106 +226,111 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
     // diagonal or the sequence end
     const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
-    float p_val[4][4];
-    __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
-#pragma unroll
-    for (int n = 0; n < 4; ++n) {
-      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int kf = 0; kf < KFRAGS; ++kf)
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            qfrag[kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK), acc, 0, 0, 0);
-      if (edge) {
-        const int kv_abs = kv0 + n * 16 + col;
-#pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          const int row_abs = q_start + wave * 16 + kgrp * 4 + reg;
-          float s = acc[reg] * scale;
-          if (kv_abs > row_abs || kv_abs >= S) s = -INFINITY;
-          p_val[n][reg] = s;
-        }
-      } else {
-#pragma unroll
-        for (int reg = 0; reg < 4; ++reg) p_val[n][reg] = acc[reg] * scale;
-      }
-    }
-    __builtin_amdgcn_s_setprio(0);
+    // stage tile kt+1 DURING compute (see the one-barrier-per-tile note in
+    // the dq kernel): its last readers finished at the barrier ending tile
+    // kt-1; placed between the two row-fragment passes' MFMA work below
+    bool staged = false;
 
-    // stage tile kt+1 into the other buffer DURING compute: its last
-    // readers finished at the barrier that ended tile kt-1, and no wave
-    // enters tile kt+1 before this tile's end barrier — one barrier per
-    // tile, LDS writes overlapped with the softmax/PV phase.  Placed after
-    // the QK^T cluster so the MFMA chain is not delayed behind staging.
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-    }
-
-    // online softmax per row (4 regs per lane)
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      float rmax = fmaxf(fmaxf(p_val[0][reg], p_val[1][reg]),
-                         fmaxf(p_val[2][reg], p_val[3][reg]));
-      rmax = rowgroup_max(rmax);
-      const float m_new = fmaxf(m_run[reg], rmax);
-      const float alpha = (m_new == -INFINITY) ? 1.f : __expf(m_run[reg] - m_new);
-      float rsum = 0.f;
+    for (int rf = 0; rf < RF; ++rf) {
+      float p_val[4][4];
+      __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
-        const float p = (m_new == -INFINITY) ? 0.f : __expf(p_val[n][reg] - m_new);
-        p_val[n][reg] = p;
-        rsum += p;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kf = 0; kf < KFRAGS; ++kf)
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qfrag[rf][kf], lds_frag(kb, n * 16 + col, kf * 32 + kgrp * 8, LDK),
+              acc, 0, 0, 0);
+        if (edge) {
+          const int kv_abs = kv0 + n * 16 + col;
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) {
+            const int row_abs = q_start + rf * 128 + wave * 16 + kgrp * 4 + reg;
+            float s = acc[reg] * scale;
+            if (kv_abs > row_abs || kv_abs >= S) s = -INFINITY;
+            p_val[n][reg] = s;
+          }
+        } else {
+#pragma unroll
+          for (int reg = 0; reg < 4; ++reg) p_val[n][reg] = acc[reg] * scale;
+        }
       }
-      rsum = rowgroup_sum(rsum);
-      l_run[reg] = l_run[reg] * alpha + rsum;
-      m_run[reg] = m_new;
-#pragma unroll
-      for (int t = 0; t < NT_HD; ++t) o_acc[t][reg] *= alpha;
-    }
+      __builtin_amdgcn_s_setprio(0);
 
-    // redistribute P (C-layout) -> A-layout via this wave's private LDS
-    // region: no block barrier, the compiler orders the wave's own
-    // ds_write -> ds_read dependency
-    __bf16* pw = lds_p + wave * 16 * LDP;
-#pragma unroll
-    for (int n = 0; n < 4; ++n)
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg)
-        pw[(kgrp * 4 + reg) * LDP + n * 16 + col] = (__bf16)p_val[n][reg];
+      if (!staged && kt + 1 < n_tiles) {
+        staged = true;
+        tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+        tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+        if (kt + 2 < n_tiles) {
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+        }
+      }
 
-    // PV: two K=32 steps over the 64-row kv tile
-    __builtin_amdgcn_s_setprio(1);
+      // online softmax per row (4 regs per lane)
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDP);
+      for (int reg = 0; reg < 4; ++reg) {
+        float rmax = fmaxf(fmaxf(p_val[0][reg], p_val[1][reg]),
+                           fmaxf(p_val[2][reg], p_val[3][reg]));
+        rmax = rowgroup_max(rmax);
+        const float m_new = fmaxf(m_run[rf][reg], rmax);
+        const float alpha = (m_new == -INFINITY) ? 1.f : __expf(m_run[rf][reg] - m_new);
+        float rsum = 0.f;
 #pragma unroll
-      for (int t = 0; t < NT_HD; ++t)
-        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a, ldsT_frag(vb, t * 16 + col, ks * 32 + kgrp * 8), o_acc[t], 0, 0, 0);
+        for (int n = 0; n < 4; ++n) {
+          const float p = (m_new == -INFINITY) ? 0.f : __expf(p_val[n][reg] - m_new);
+          p_val[n][reg] = p;
+          rsum += p;
+        }
+        rsum = rowgroup_sum(rsum);
+        l_run[rf][reg] = l_run[rf][reg] * alpha + rsum;
+        m_run[rf][reg] = m_new;
+#pragma unroll
+        for (int t = 0; t < NT_HD; ++t) o_acc[rf][t][reg] *= alpha;
+      }
+
+      // P relayout via this (wave, rf)'s private LDS region (no barrier)
+      __bf16* pw = lds_p + (rf * 8 + wave) * 16 * LDP;
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg)
+          pw[(kgrp * 4 + reg) * LDP + n * 16 + col] = (__bf16)p_val[n][reg];
+
+      // PV: two K=32 steps over the 64-row kv tile
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const bf16x8 a = lds_frag(pw, col, ks * 32 + kgrp * 8, LDP);
+#pragma unroll
+        for (int t = 0; t < NT_HD; ++t)
+          o_acc[rf][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a, ldsT_frag(vb, t * 16 + col, ks * 32 + kgrp * 8), o_acc[rf][t], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
   // epilogue: O = o_acc / l, LSE = m + log(l)
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    const int row_abs = q_start + wave * 16 + kgrp * 4 + reg;
-    if (row_abs >= S) continue;
-    const float inv_l = (l_run[reg] > 0.f) ? 1.f / l_run[reg] : 0.f;
+  for (int rf = 0; rf < RF; ++rf)
 #pragma unroll
-    for (int t = 0; t < NT_HD; ++t) {
-      const int c = t * 16 + col;
-      if (c < hd)
-        out[base + (long)row_abs * hd + c] = __float2bfloat16(o_acc[t][reg] * inv_l);
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row_abs = q_start + rf * 128 + wave * 16 + kgrp * 4 + reg;
+      if (row_abs >= S) continue;
+      const float inv_l = (l_run[rf][reg] > 0.f) ? 1.f / l_run[rf][reg] : 0.f;
+#pragma unroll
+      for (int t = 0; t < NT_HD; ++t) {
+        const int c = t * 16 + col;
+        if (c < hd)
+          out[base + (long)row_abs * hd + c] = __float2bfloat16(o_acc[rf][t][reg] * inv_l);
+      }
+      if (col == 0)
+        lse[(long)bh * S + row_abs] = m_run[rf][reg] + __logf(fmaxf(l_run[rf][reg], 1e-30f));
     }
-    if (col == 0)
-      lse[(long)bh * S + row_abs] = m_run[reg] + __logf(fmaxf(l_run[reg], 1e-30f));
-  }
 }
 
 // ---------------------------------------------------------------------------
@@ -688,10 +706,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  dim3 grid((S + 127) / 128, B * nh), block(512);
+  dim3 grid((S + 255) / 256, B * nh), block(512);
   DISPATCH_HD(HDP, {
     const int LDK = HD + LPAD, LDP = TILE + LPAD;
-    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + 8 * 16 * LDP) * sizeof(__bf16);
+    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + 2 * 8 * 16 * LDP) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
