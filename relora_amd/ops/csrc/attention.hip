@@ -156,9 +156,11 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   constexpr int LDK = HD + LPAD;
   constexpr int LDP = TILE + LPAD;
   constexpr int NV = (HD + 63) / 64;   // bf16x8 staging slices per thread
-  constexpr int RF = 2;                // q row-fragments per wave: 256-row
-                                       // blocks halve K/V traffic + staging
-                                       // per MFMA vs 128-row blocks
+  // q row-fragments per wave: RF=2 (256-row blocks) halves K/V traffic and
+  // staging per MFMA, but its register footprint costs hd64 its second
+  // resident block (2 -> 1 block/CU, measured -20%); at hd>=96 residency is
+  // 1 block/CU either way and RF=2 measured +25%
+  constexpr int RF = (HD <= 64) ? 1 : 2;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_k = (__bf16*)smem;              // [2][TILE][LDK]
@@ -706,10 +708,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto out = torch::empty_like(q);
   auto lse = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  dim3 grid((S + 255) / 256, B * nh), block(512);
+  dim3 block(512);
   DISPATCH_HD(HDP, {
+    constexpr int RF = (HD <= 64) ? 1 : 2;
+    dim3 grid((S + 128 * RF - 1) / (128 * RF), B * nh);
     const int LDK = HD + LPAD, LDP = TILE + LPAD;
-    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + 2 * 8 * 16 * LDP) * sizeof(__bf16);
+    size_t smem = (2 * TILE * LDK + 2 * HD * TILE + RF * 8 * 16 * LDP) * sizeof(__bf16);
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
