@@ -304,10 +304,12 @@ void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
 //   dB^T        = skinny_grad(P=t_u,  X=dy) -> [r, N]
 // ---------------------------------------------------------------------------
 
-// grid: (C/128, MCHUNKS); block 256 (4 waves); out tile [r<=128][128]
+// grid: (C/128, MCHUNKS, r/128); block 256 (4 waves); out tile [128][128]
 __global__ __launch_bounds__(256) void skinny_grad_kernel(
     const __hip_bfloat16* __restrict__ P, const __hip_bfloat16* __restrict__ X,
     float* __restrict__ part, long M, int C, int r, int rows_per_chunk) {
+  const int r0 = blockIdx.z * 128;      // r-tile (rank 256 spans two)
+  const int rtile = min(r - r0, 128);
   constexpr int LDT = 64 + LPAD;  // transposed tile row stride (m dim)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* pt = (__bf16*)smem;        // [r][LDT]   P^T tile (m contiguous)
@@ -330,13 +332,13 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
     for (int j = 0; j < 8; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   for (long m0 = m_begin; m0 < m_end; m0 += 64) {
-    // stage P^T: pt[j][mm] = P[m0+mm][j]; threads cover 64 rows x r/8 vec8
-    for (int t = threadIdx.x; t < 64 * (r / 8); t += blockDim.x) {
-      const int mm = t / (r / 8);
-      const int j8 = (t % (r / 8)) * 8;
+    // stage P^T slice: pt[j][mm] = P[m0+mm][r0+j]
+    for (int t = threadIdx.x; t < 64 * (rtile / 8); t += blockDim.x) {
+      const int mm = t / (rtile / 8);
+      const int j8 = (t % (rtile / 8)) * 8;
       bf16x8 v;
       if (m0 + mm < m_end) {
-        v = *reinterpret_cast<const bf16x8*>(P + (m0 + mm) * (long)r + j8);
+        v = *reinterpret_cast<const bf16x8*>(P + (m0 + mm) * (long)r + r0 + j8);
       } else {
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
@@ -368,7 +370,7 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
         const int jrow = wave * 32 + i * 16 + col;
-        const bf16x8 a = (jrow < r)
+        const bf16x8 a = (jrow < rtile)
             ? *reinterpret_cast<const bf16x8*>(pt + jrow * LDT + ks * 32 + kgrp * 8)
             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
@@ -394,7 +396,7 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
         const int rr = rbase + reg;
-        if (rr < r) plane[(long)rr * C + c] = acc[i][j][reg];
+        if (rr < rtile) plane[(long)(r0 + rr) * C + c] = acc[i][j][reg];
       }
     }
   }
@@ -429,7 +431,7 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
   const long M = P.size(0);
   const int r = P.size(1);
   const int C = X.size(1);
-  TORCH_CHECK(X.size(0) == M && r % 8 == 0 && r <= 128);
+  TORCH_CHECK(X.size(0) == M && r % 8 == 0 && r <= 256);
   // chunk M so the grid fills the chip: (C/128)*chunks >= ~512
   int chunks = 1;
   const int ctiles = (C + 127) / 128;
@@ -442,7 +444,7 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   constexpr int LDT = 64 + LPAD;
   const size_t lds = (128 * LDT + 128 * LDT) * sizeof(__bf16);
-  dim3 grid(ctiles, chunks), block(256);
+  dim3 grid(ctiles, chunks, (r + 127) / 128), block(256);
   hipLaunchKernelGGL(skinny_grad_kernel, grid, block, lds, stream,
                      (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
                      part.data_ptr<float>(), M, C, r, rows);

@@ -163,7 +163,8 @@ def test_relora_linear_uses_fused_path_on_gpu():
 
 def test_skinny_grad_matches_matmul():
     torch.manual_seed(5)
-    for M, r, C in ((4096, 128, 2048), (16384, 128, 512), (1000, 64, 320)):
+    for M, r, C in ((4096, 128, 2048), (16384, 128, 512), (1000, 64, 320),
+                    (2048, 256, 1024)):
         P = torch.randn(M, r, device="cuda", dtype=torch.bfloat16) * 0.1
         X = torch.randn(M, C, device="cuda", dtype=torch.bfloat16) * 0.1
         got = ext().skinny_grad(P, X, 1.0, False, torch.float32)
