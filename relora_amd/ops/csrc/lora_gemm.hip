@@ -28,6 +28,17 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define LPAD 8  // bf16 elements of LDS row padding (one 16B slot)
 
+// Rotated layout for transposed LDS tiles with 64-element rows: element
+// (row, c) of an [R][64] image lives at row*64 + rot*8 + (c&7), with
+// rot = ((c>>3) + (row>>3) + (row&7)) & 7.  Exhaustive-search layout (see
+// attention.hip): both 8-scalar transpose writes (8 consecutive rows OR 8
+// consecutive c) and bf16x8 fragment reads are bank-conflict-free, and no
+// row padding is needed.
+DEV_INLINE int rot8(int row, int c64) {
+  return ((((c64 >> 3) + (row >> 3) + (row & 7)) & 7) << 3) + (c64 & 7);
+}
+DEV_INLINE int tr64(int row, int c) { return row * 64 + rot8(row, c); }
+
 // ---------------------------------------------------------------------------
 // philox4x32-10 — counter-based RNG for the dropout mask (regenerable, but we
 // persist packed bits: exact replay in backward with zero recompute).
@@ -136,12 +147,22 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
       *reinterpret_cast<bf16x8*>(q_im + n * ldt + c) = v;
     }
   } else {
-    // Q is [r][N]: q_im[n][k] = Q[k][n0+n].  Coalesced across threads in n.
-    for (int t = threadIdx.x; t < 128 * r; t += blockDim.x) {
-      const int k = t / 128;
-      const int n = t % 128;
-      q_im[n * ldt + k] =
-          (n0 + n < N) ? (__bf16)Q[(long)k * ldq + n0 + n] : (__bf16)0.f;
+    // Q is [r][N]: q_im[n][k] = Q[k][n0+n].  bf16x8 loads along n (contiguous
+    // in global), 8 rotated scalar LDS writes each (conflict-free, see tr64)
+    for (int t = threadIdx.x; t < r * 16; t += blockDim.x) {
+      const int k = t / 16;
+      const int nb = (t % 16) * 8;
+      bf16x8 v;
+      if (n0 + nb + 8 <= N) {
+        v = *reinterpret_cast<const bf16x8*>(Q + (long)k * ldq + n0 + nb);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v[j] = (n0 + nb + j < N) ? (__bf16)Q[(long)k * ldq + n0 + nb + j] : (__bf16)0.f;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        q_im[(nb + j) * ldt + (k & ~63) + rot8(nb + j, k & 63)] = v[j];
     }
   }
   __syncthreads();
@@ -166,8 +187,11 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
           *reinterpret_cast<const bf16x8*>(p_im + (wr + mi * 16 + fr) * ldt + kk + kg);
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
-        const bf16x8 b =
-            *reinterpret_cast<const bf16x8*>(q_im + (wc + ni * 16 + fr) * ldt + kk + kg);
+        const int qrow = wc + ni * 16 + fr;
+        const int koff = kk + kg;
+        const bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            TRANSQ ? q_im + qrow * ldt + (koff & ~63) + rot8(qrow, koff & 63)
+                   : q_im + qrow * ldt + koff);
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mi][ni], 0, 0, 0);
       }
     }
@@ -310,10 +334,9 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
     float* __restrict__ part, long M, int C, int r, int rows_per_chunk) {
   const int r0 = blockIdx.z * 128;      // r-tile (rank 256 spans two)
   const int rtile = min(r - r0, 128);
-  constexpr int LDT = 64 + LPAD;  // transposed tile row stride (m dim)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* pt = (__bf16*)smem;        // [r][LDT]   P^T tile (m contiguous)
-  __bf16* xt = pt + 128 * LDT;       // [128][LDT] X^T tile (c rows, m cols)
+  __bf16* pt = (__bf16*)smem;        // [128][64] rotated: P^T tile (m contiguous)
+  __bf16* xt = pt + 128 * 64;        // [128][64] rotated: X^T tile
 
   const int c0 = blockIdx.x * 128;
   const long m_begin = (long)blockIdx.y * rows_per_chunk;
@@ -343,7 +366,7 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) pt[(j8 + j) * LDT + mm] = v[j];
+      for (int j = 0; j < 8; ++j) pt[tr64(j8 + j, mm)] = v[j];
     }
     // stage X^T: xt[cc][mm] = X[m0+mm][c0+cc]; vec8 over the row direction
     for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
@@ -361,7 +384,7 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xt[(c8 + j) * LDT + mm] = v[j];
+      for (int j = 0; j < 8; ++j) xt[tr64(c8 + j, mm)] = v[j];
     }
     __syncthreads();
 
@@ -371,12 +394,12 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
       for (int i = 0; i < 2; ++i) {
         const int jrow = wave * 32 + i * 16 + col;
         const bf16x8 a = (jrow < rtile)
-            ? *reinterpret_cast<const bf16x8*>(pt + jrow * LDT + ks * 32 + kgrp * 8)
+            ? *reinterpret_cast<const bf16x8*>(pt + tr64(jrow, ks * 32 + kgrp * 8))
             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-              xt + (j * 16 + col) * LDT + ks * 32 + kgrp * 8);
+              xt + tr64(j * 16 + col, ks * 32 + kgrp * 8));
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i][j], 0, 0, 0);
         }
       }
@@ -442,8 +465,7 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
   chunks = (int)((M + rows - 1) / rows);
   auto part = torch::empty({chunks, r, C}, P.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  constexpr int LDT = 64 + LPAD;
-  const size_t lds = (128 * LDT + 128 * LDT) * sizeof(__bf16);
+  const size_t lds = (2u * 128 * 64) * sizeof(__bf16);
   dim3 grid(ctiles, chunks, (r + 127) / 128), block(256);
   hipLaunchKernelGGL(skinny_grad_kernel, grid, block, lds, stream,
                      (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
