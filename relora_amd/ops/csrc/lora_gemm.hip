@@ -335,8 +335,10 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
   const int r0 = blockIdx.z * 128;      // r-tile (rank 256 spans two)
   const int rtile = min(r - r0, 128);
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  __bf16* pt = (__bf16*)smem;        // [128][64] rotated: P^T tile (m contiguous)
-  __bf16* xt = pt + 128 * 64;        // [128][64] rotated: X^T tile
+  // double-buffered rotated tiles; staging for m-tile t+1 overlaps the MFMA
+  // work of tile t (one barrier per m-tile — same scheme as attention.hip)
+  __bf16* pt = (__bf16*)smem;        // [2][128][64] rotated: P^T tiles
+  __bf16* xt = pt + 2 * 128 * 64;    // [2][128][64] rotated: X^T tiles
 
   const int c0 = blockIdx.x * 128;
   const long m_begin = (long)blockIdx.y * rows_per_chunk;
@@ -354,8 +356,10 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (long m0 = m_begin; m0 < m_end; m0 += 64) {
-    // stage P^T slice: pt[j][mm] = P[m0+mm][r0+j]
+  auto stage = [&](long m0, int buf) {
+    __bf16* ptb = pt + buf * 128 * 64;
+    __bf16* xtb = xt + buf * 128 * 64;
+    // P^T slice: pt[j][mm] = P[m0+mm][r0+j]
     for (int t = threadIdx.x; t < 64 * (rtile / 8); t += blockDim.x) {
       const int mm = t / (rtile / 8);
       const int j8 = (t % (rtile / 8)) * 8;
@@ -366,9 +370,9 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) pt[tr64(j8 + j, mm)] = v[j];
+      for (int j = 0; j < 8; ++j) ptb[tr64(j8 + j, mm)] = v[j];
     }
-    // stage X^T: xt[cc][mm] = X[m0+mm][c0+cc]; vec8 over the row direction
+    // X^T: xt[cc][mm] = X[m0+mm][c0+cc]; vec8 over the row direction
     for (int t = threadIdx.x; t < 64 * 16; t += blockDim.x) {
       const int mm = t / 16;
       const int c8 = (t % 16) * 8;
@@ -384,22 +388,29 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
 #pragma unroll
-      for (int j = 0; j < 8; ++j) xt[tr64(c8 + j, mm)] = v[j];
+      for (int j = 0; j < 8; ++j) xtb[tr64(c8 + j, mm)] = v[j];
     }
-    __syncthreads();
+  };
 
+  stage(m_begin, 0);
+  __syncthreads();
+  for (long m0 = m_begin; m0 < m_end; m0 += 64) {
+    const int buf = (int)(((m0 - m_begin) >> 6) & 1);
+    if (m0 + 64 < m_end) stage(m0 + 64, buf ^ 1);
+    const __bf16* ptb = pt + buf * 128 * 64;
+    const __bf16* xtb = xt + buf * 128 * 64;
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
         const int jrow = wave * 32 + i * 16 + col;
         const bf16x8 a = (jrow < rtile)
-            ? *reinterpret_cast<const bf16x8*>(pt + tr64(jrow, ks * 32 + kgrp * 8))
+            ? *reinterpret_cast<const bf16x8*>(ptb + tr64(jrow, ks * 32 + kgrp * 8))
             : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-              xt + tr64(j * 16 + col, ks * 32 + kgrp * 8));
+              xtb + tr64(j * 16 + col, ks * 32 + kgrp * 8));
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i][j], 0, 0, 0);
         }
       }
@@ -465,7 +476,7 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
   chunks = (int)((M + rows - 1) / rows);
   auto part = torch::empty({chunks, r, C}, P.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  const size_t lds = (2u * 128 * 64) * sizeof(__bf16);
+  const size_t lds = (4u * 128 * 64) * sizeof(__bf16);
   dim3 grid(ctiles, chunks, (r + 127) / 128), block(256);
   hipLaunchKernelGGL(skinny_grad_kernel, grid, block, lds, stream,
                      (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
