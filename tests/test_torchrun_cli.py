@@ -13,9 +13,11 @@ import pytest
 @pytest.mark.timeout(600)
 def test_torchrun_two_proc_cli(tmp_path):
     env = dict(os.environ)
-    env.pop("RANK", None)
-    env.pop("WORLD_SIZE", None)
-    env.pop("LOCAL_RANK", None)
+    # a prior in-process test may have exported MASTER_* while its (destroyed)
+    # TCPStore still listens on that port; torchrun's children would rendezvous
+    # against the stale store and hang — strip all launcher state
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
+        env.pop(k, None)
     env["RELORA_AMD_NO_TQDM"] = "1"
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
