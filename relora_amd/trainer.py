@@ -322,7 +322,8 @@ def main(args):
 
     if global_rank == 0:
         wandb.init(project="peft_pretraining", tags=args.tags, id=wandb_id,
-                   resume="allow", notes=args.comment)
+                   resume="allow", notes=args.comment,
+                   dir=args.save_dir if args.save_dir else None)
         args.run_name = wandb.run.name
         if args.save_dir is None:
             args.save_dir = f"checkpoints/{wandb.run.name}"
