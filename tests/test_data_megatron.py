@@ -305,3 +305,35 @@ def test_args_json_sidecar_roundtrip(tmp_path):
     with open(p) as f:
         d = yaml.safe_load(f)
     assert json.loads(json.dumps(d)) == d
+
+
+def test_tokenize_and_chunk_offline(tmp_path):
+    """HF-path preprocessing (dataloader.py tokenize_and_chunk) with an
+    offline word-level tokenizer: concat + chunk to block_size, EOS joined,
+    attention_mask dropped (reference dataloader.py:57-124)."""
+    import datasets as hfds
+    from tokenizers import Tokenizer as Tk, models as tkm, pre_tokenizers as tkp
+    from transformers import PreTrainedTokenizerFast
+
+    from relora_amd.data.dataloader import tokenize_and_chunk
+
+    vocab = {"[PAD]": 0, "[UNK]": 1, "</s>": 2}
+    for w in ("aa", "bb", "cc", "dd", "ee"):
+        vocab[w] = len(vocab)
+    tk = Tk(tkm.WordLevel(vocab=vocab, unk_token="[UNK]"))
+    tk.pre_tokenizer = tkp.Whitespace()
+    tok = PreTrainedTokenizerFast(tokenizer_object=tk, pad_token="[PAD]",
+                                  unk_token="[UNK]", eos_token="</s>")
+
+    ds = hfds.DatasetDict({"train": hfds.Dataset.from_dict(
+        {"text": ["aa bb cc", "dd ee", "aa aa aa aa aa"]})})
+    out = tokenize_and_chunk(tok, ds, text_field="text", sequence_length=4,
+                             num_cpu=1)["train"]
+    assert out.column_names == ["input_ids"]
+    rows = [r["input_ids"] for r in out]
+    assert all(len(r) == 4 for r in rows)
+    # total tokens = sum(doc tokens + 1 eos each), floored to blocks of 4
+    total = sum(len(tok(t)["input_ids"]) + 1 for t in ["aa bb cc", "dd ee", "aa aa aa aa aa"])
+    assert len(rows) == total // 4
+    flat = [t for r in rows for t in r]
+    assert tok.eos_token_id in flat
