@@ -21,8 +21,8 @@ def test_torchrun_two_proc_cli(tmp_path):
     env["RELORA_AMD_NO_TQDM"] = "1"
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
+        "--standalone", "--local-addr", "127.0.0.1",
         "--nnodes=1", "--nproc-per-node=2",
-        "--master-addr", "127.0.0.1", "--master-port", "29731",
         "torchrun_main.py",
         "--model_config", "configs/llama_9m.json",
         "--synthetic_data", "true",
