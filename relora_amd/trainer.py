@@ -779,7 +779,16 @@ def main(args):
         pbar.close()
 
     current_model_directory = f"{args.save_dir}/model_{update_step}"
-    if not os.path.exists(current_model_directory):
+    # The reference guards the final save with a per-rank os.path.exists
+    # (torchrun_main.py:956) — a cross-rank race: rank 0 creates the
+    # directory inside save_model while a slower rank then sees it existing,
+    # skips the save (and its barrier), and the job deadlocks between
+    # save_model's barrier and the final eval's all_reduce.  Decide once on
+    # rank 0 and broadcast so every rank takes the same branch.
+    need_final_save = [not os.path.exists(current_model_directory)]
+    if dist.is_initialized() and world_size > 1:
+        dist.broadcast_object_list(need_final_save, src=0)
+    if need_final_save[0]:
         training_state_checkpoint = {
             "global_step": global_step,
             "update_step": update_step,

@@ -12,13 +12,16 @@ import pytest
 
 @pytest.mark.timeout(600)
 def test_torchrun_two_proc_cli(tmp_path):
-    env = dict(os.environ)
-    # a prior in-process test may have exported MASTER_* while its (destroyed)
-    # TCPStore still listens on that port; torchrun's children would rendezvous
-    # against the stale store and hang — strip all launcher state
-    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR", "MASTER_PORT"):
-        env.pop(k, None)
+    # hermetic environment: prior in-process tests leave launcher/rendezvous
+    # state (MASTER_*, TCPStore listeners, gloo sockets) that can poison the
+    # child rendezvous — start from a minimal env instead of inheriting
+    keep = ("PATH", "HOME", "TMPDIR", "LD_LIBRARY_PATH", "ROCM_PATH",
+            "HSA_ENABLE_IPC_MODE_LEGACY", "PYTHONPATH", "HIP_VISIBLE_DEVICES")
+    env = {k: os.environ[k] for k in keep if k in os.environ}
     env["RELORA_AMD_NO_TQDM"] = "1"
+    env["OMP_NUM_THREADS"] = "1"
+    if os.environ.get("RELORA_AMD_HANG_DUMP_S"):
+        env["RELORA_AMD_HANG_DUMP_S"] = os.environ["RELORA_AMD_HANG_DUMP_S"]
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--standalone", "--local-addr", "127.0.0.1",
@@ -36,9 +39,15 @@ def test_torchrun_two_proc_cli(tmp_path):
         "--eval_every", "100", "--save_every", "100", "--workers", "0",
         "--save_dir", str(tmp_path / "run"),
     ]
-    res = subprocess.run(cmd, cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
-                         env=env, capture_output=True, text=True, timeout=560)
-    assert res.returncode == 0, res.stdout[-2000:] + res.stderr[-2000:]
+    # write through real files, detached from pytest's captured fds (the
+    # piped form hangs under pytest's fd-level capture)
+    out_path = tmp_path / "torchrun.out"
+    with open(out_path, "w") as out:
+        res = subprocess.run(
+            cmd, cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            env=env, stdin=subprocess.DEVNULL, stdout=out,
+            stderr=subprocess.STDOUT, timeout=560, start_new_session=True)
+    assert res.returncode == 0, out_path.read_text()[-2000:]
     state = json.load(open(tmp_path / "run" / "model_4" / "training_state.json"))
     assert state["update_step"] == 4
     # tokens counted x world_size: 4 update steps x 8 x 32 tokens

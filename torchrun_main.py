@@ -8,8 +8,14 @@ or with a yaml recipe:
     torchrun --nproc-per-node 8 torchrun_main.py --training_config training_configs/1B_v1.0.yaml
 """
 
+import os
+
 from relora_amd.trainer import main, parse_args
 
 if __name__ == "__main__":
+    if os.environ.get("RELORA_AMD_HANG_DUMP_S"):
+        import faulthandler
+        faulthandler.dump_traceback_later(
+            int(os.environ["RELORA_AMD_HANG_DUMP_S"]), repeat=True)
     args = parse_args()
     main(args)
