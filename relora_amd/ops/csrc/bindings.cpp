@@ -34,7 +34,8 @@ std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t s
 void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q);
 void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
                   torch::Tensor mask, double inv_keep);
-torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
+torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, torch::Tensor xmask,
+                          double inv_keep, double scale,
                           bool transpose_out, torch::ScalarType dtype);
 // quantize.hip
 std::vector<torch::Tensor> quantize_nf4(torch::Tensor x);

@@ -328,9 +328,13 @@ void lora_add_nn_(torch::Tensor out, torch::Tensor P, torch::Tensor Q,
 //   dB^T        = skinny_grad(P=t_u,  X=dy) -> [r, N]
 // ---------------------------------------------------------------------------
 
-// grid: (C/128, MCHUNKS, r/128); block 256 (4 waves); out tile [128][128]
+// grid: (C/128, MCHUNKS, r/128); block 256 (4 waves); out tile [128][128].
+// MASK: X is consumed as dropout(X) = maskbit * X * inv_keep, applied while
+// staging (so the forward never has to persist the dropped-out activations).
+template <bool MASK>
 __global__ __launch_bounds__(256) void skinny_grad_kernel(
     const __hip_bfloat16* __restrict__ P, const __hip_bfloat16* __restrict__ X,
+    const uint8_t* __restrict__ xmask, float inv_keep,
     float* __restrict__ part, long M, int C, int r, int rows_per_chunk) {
   const int r0 = blockIdx.z * 128;      // r-tile (rank 256 spans two)
   const int rtile = min(r - r0, 128);
@@ -380,10 +384,25 @@ __global__ __launch_bounds__(256) void skinny_grad_kernel(
       const long xoff = (m0 + mm) * (long)C + c0 + c8;
       if (m0 + mm < m_end && c0 + c8 + 8 <= C && (xoff & 7) == 0) {
         v = *reinterpret_cast<const bf16x8*>(X + xoff);
+        if (MASK) {
+          // flat-packed bits; an aligned vec8 covers bits j..j+7 of bytes
+          // (xoff>>3) and possibly (xoff>>3)+1 when xoff&7 != 0 — here
+          // (xoff&7)==0 so exactly one byte
+          const uint8_t mb = xmask[xoff >> 3];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            v[j] = (mb >> j) & 1 ? (__bf16)((float)v[j] * inv_keep) : (__bf16)0.f;
+        }
       } else if (m0 + mm < m_end) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          v[j] = (c0 + c8 + j < C) ? (__bf16)X[xoff + j] : (__bf16)0.f;
+        for (int j = 0; j < 8; ++j) {
+          float f = (c0 + c8 + j < C) ? (float)(__bf16)X[xoff + j] : 0.f;
+          if (MASK && c0 + c8 + j < C) {
+            const uint8_t mb = xmask[(xoff + j) >> 3];
+            f = (mb >> ((xoff + j) & 7)) & 1 ? f * inv_keep : 0.f;
+          }
+          v[j] = (__bf16)f;
+        }
       } else {
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
@@ -457,8 +476,10 @@ __global__ void skinny_combine_kernel(const float* __restrict__ part,
   }
 }
 
-// out = scale * (P[M,r]^T @ X[M,C]) in `dtype`; [C, r] when transpose_out
-torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
+// out = scale * (P[M,r]^T @ dropout_mask(X)[M,C]) in `dtype`; [C, r] when
+// transpose_out.  `xmask` empty -> X used as-is.
+torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, torch::Tensor xmask,
+                          double inv_keep, double scale,
                           bool transpose_out, torch::ScalarType dtype) {
   TORCH_CHECK(P.is_cuda() && P.is_contiguous() && X.is_contiguous());
   TORCH_CHECK(P.scalar_type() == torch::kBFloat16 && X.scalar_type() == torch::kBFloat16);
@@ -478,9 +499,20 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, double scale,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   const size_t lds = (4u * 128 * 64) * sizeof(__bf16);
   dim3 grid(ctiles, chunks, (r + 127) / 128), block(256);
-  hipLaunchKernelGGL(skinny_grad_kernel, grid, block, lds, stream,
-                     (const __hip_bfloat16*)P.data_ptr(), (const __hip_bfloat16*)X.data_ptr(),
-                     part.data_ptr<float>(), M, C, r, rows);
+  const bool has_mask = xmask.defined() && xmask.numel() > 0;
+  if (has_mask) {
+    TORCH_CHECK(xmask.numel() == (M * (long)C + 7) / 8, "skinny_grad mask size");
+    hipLaunchKernelGGL((skinny_grad_kernel<true>), grid, block, lds, stream,
+                       (const __hip_bfloat16*)P.data_ptr(),
+                       (const __hip_bfloat16*)X.data_ptr(),
+                       xmask.data_ptr<uint8_t>(), (float)inv_keep,
+                       part.data_ptr<float>(), M, C, r, rows);
+  } else {
+    hipLaunchKernelGGL((skinny_grad_kernel<false>), grid, block, lds, stream,
+                       (const __hip_bfloat16*)P.data_ptr(),
+                       (const __hip_bfloat16*)X.data_ptr(), nullptr, 1.f,
+                       part.data_ptr<float>(), M, C, r, rows);
+  }
   HIP_CHECK_LAST();
   auto out = transpose_out
       ? torch::empty({C, r}, P.options().dtype(dtype))

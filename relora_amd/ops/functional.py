@@ -362,10 +362,12 @@ class _FusedLoRALinear(torch.autograd.Function):
         else:
             xd, mask = x2d, None
         t_u = xd @ lora_A.t()                       # [M, r]
+        del xd  # dA re-applies the mask inline in skinny_grad; no need to
+                # persist the dropped-out copy (64 MB per flagship linear)
         bs = lora_B * scale                         # [N, r]
         y = F.linear(x2d, weight, bias)
         hip.ext().lora_add_nt_(y, t_u, bs)          # y += t_u @ bs^T
-        ctx.save_for_backward(x2d, xd, mask if mask is not None else x2d.new_empty(0),
+        ctx.save_for_backward(x2d, mask if mask is not None else x2d.new_empty(0),
                               t_u, weight, lora_A, bs)
         ctx.scale = scale
         ctx.dropout_p = dropout_p if use_dropout else 0.0
@@ -374,7 +376,7 @@ class _FusedLoRALinear(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x2d, xd, mask, t_u, weight, lora_A, bs = ctx.saved_tensors
+        x2d, mask, t_u, weight, lora_A, bs = ctx.saved_tensors
         scale, p = ctx.scale, ctx.dropout_p
         N = weight.shape[0]
         dy2d = dy.contiguous().view(-1, N)
@@ -385,11 +387,15 @@ class _FusedLoRALinear(torch.autograd.Function):
                                mask if p > 0 else mask.new_empty(0, dtype=torch.uint8),
                                1.0 / (1.0 - p) if p > 0 else 1.0)
 
-        # dA = u_s^T @ xd, dB = s * dy^T @ t_u = (t_u^T @ dy)^T * s — via the
-        # chunked-M skinny_grad kernel (hipBLASLt launches these 32-WG wide);
-        # scale/transpose/cast fold into its combine stage
-        dA = hip.ext().skinny_grad(u_s, xd, 1.0, False, lora_A.dtype)
-        dB = hip.ext().skinny_grad(t_u, dy2d, scale, True, bs.dtype)
+        # dA = u_s^T @ dropout(x), dB = s * dy^T @ t_u = (t_u^T @ dy)^T * s —
+        # via the chunked-M skinny_grad kernel (hipBLASLt launches these
+        # 32-WG wide); the dropout mask re-applies inline while staging x,
+        # and scale/transpose/cast fold into the combine stage
+        empty_mask = mask.new_empty(0, dtype=torch.uint8)
+        dA = hip.ext().skinny_grad(u_s, x2d, mask if p > 0 else empty_mask,
+                                   1.0 / (1.0 - p) if p > 0 else 1.0,
+                                   1.0, False, lora_A.dtype)
+        dB = hip.ext().skinny_grad(t_u, dy2d, empty_mask, 1.0, scale, True, bs.dtype)
         dw = dy2d.t() @ x2d if ctx.needs_input_grad[1] else None
         dbias = dy2d.sum(0) if ctx.has_bias and ctx.needs_input_grad[2] else None
         return (dx.view(dy.shape[:-1] + (weight.shape[1],)), dw, dbias,

@@ -167,14 +167,15 @@ def test_skinny_grad_matches_matmul():
                     (2048, 256, 1024)):
         P = torch.randn(M, r, device="cuda", dtype=torch.bfloat16) * 0.1
         X = torch.randn(M, C, device="cuda", dtype=torch.bfloat16) * 0.1
-        got = ext().skinny_grad(P, X, 1.0, False, torch.float32)
+        nomask = torch.empty(0, device="cuda", dtype=torch.uint8)
+        got = ext().skinny_grad(P, X, nomask, 1.0, 1.0, False, torch.float32)
         ref = P.float().t() @ X.float()
         # bf16 products, fp32 accumulation: tolerance scales with sqrt(M)
         tol = 0.03 * ref.abs().max().item() + 0.05
         err = (got - ref).abs().max().item()
         assert err < tol, (M, r, C, err, tol)
         # transposed + scaled + bf16 variant
-        gt = ext().skinny_grad(P, X, 0.5, True, torch.bfloat16)
+        gt = ext().skinny_grad(P, X, nomask, 1.0, 0.5, True, torch.bfloat16)
         assert gt.shape == (C, r)
         errt = (gt.float() - 0.5 * ref.t()).abs().max().item()
         assert errt < tol, (M, r, C, errt)
@@ -265,3 +266,16 @@ def test_fused_lora_linear_odd_dims():
         y3.backward(torch.randn_like(y3))
         frac = (x3.grad == 0).float().mean().item()
         assert abs(frac - 0.3) < 0.06, frac
+
+
+def test_skinny_grad_masked_matches_oracle():
+    """dA with the dropout mask applied inline while staging x."""
+    torch.manual_seed(9)
+    M, r, C, p = 2048, 128, 1024, 0.3
+    P = torch.randn(M, r, device="cuda", dtype=torch.bfloat16) * 0.1
+    x = torch.randn(M, C, device="cuda", dtype=torch.bfloat16)
+    xd, mask = ext().dropout_mask_fwd(x, p, 11)
+    got = ext().skinny_grad(P, x, mask, 1.0 / (1.0 - p), 1.0, False, torch.float32)
+    ref = P.float().t() @ xd.float()
+    tol = 0.03 * ref.abs().max().item() + 0.05
+    assert (got - ref).abs().max().item() < tol
