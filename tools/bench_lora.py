@@ -41,9 +41,10 @@ def main():
     print(f"lora_add_nt   : {d*1e6:8.1f} us  ({(2*M*N*2)/d/1e9:6.0f} GB/s eff)")
     d = t(lambda: ext.lora_add_nn_(dx, us, A, mask, 1.0 / 0.9))
     print(f"lora_add_nn   : {d*1e6:8.1f} us  ({(2*M*K*2)/d/1e9:6.0f} GB/s eff)")
-    d = t(lambda: ext.skinny_grad(us, xd, 1.0, False, torch.bfloat16))
-    print(f"skinny_grad dA: {d*1e6:8.1f} us  ({(M*K*2)/d/1e9:6.0f} GB/s eff)")
-    d = t(lambda: ext.skinny_grad(tu, dy, 0.25, True, torch.bfloat16))
+    nomask = torch.empty(0, device="cuda", dtype=torch.uint8)
+    d = t(lambda: ext.skinny_grad(us, x, mask, 1.0 / 0.9, 1.0, False, torch.bfloat16))
+    print(f"skinny_grad dA: {d*1e6:8.1f} us  ({(M*K*2)/d/1e9:6.0f} GB/s eff, masked)")
+    d = t(lambda: ext.skinny_grad(tu, dy, nomask, 1.0, 0.25, True, torch.bfloat16))
     print(f"skinny_grad dB: {d*1e6:8.1f} us  ({(M*N*2)/d/1e9:6.0f} GB/s eff)")
 
 
