@@ -258,3 +258,19 @@ def test_auto_micro_batch_estimator():
     assert b > a
     # a tiny budget degrades gracefully to 1
     assert auto_micro_batch(cfg, 2048, 1024, 8, hbm_bytes=1 << 30) == 1
+
+
+def test_training_config_yaml_recipes_parse(tmp_path):
+    """The shipped training_configs/ recipes must parse through the YAML
+    override path (reference args_utils.py:9-21 semantics)."""
+    import yaml as _yaml
+
+    for recipe in ("training_configs/1B_v1.0.yaml", "training_configs/250M_v1.0.yaml"):
+        with open(recipe) as f:
+            overrides = _yaml.safe_load(f)
+        args = parse_args(["--training_config", recipe])
+        assert args.use_peft is True
+        assert args.lora_r == overrides["lora_r"]
+        assert args.total_batch_size == overrides["total_batch_size"]
+        assert float(args.lr) == float(overrides["lr"])
+        assert args.num_training_steps == int(str(overrides["num_training_steps"]).replace("_", ""))
