@@ -77,8 +77,11 @@ class DistributedModel(nn.Module):
         if _dist_active() and broadcast_params:
             with torch.no_grad():
                 for t in list(module.parameters()) + list(module.buffers()):
-                    if t.is_floating_point() or t.dtype in (torch.int64, torch.int32):
-                        dist.broadcast(t.data, src=0, group=self.process_group)
+                    # everything replicable: floats, ints, and the uint8/int8
+                    # payloads of quantized frozen weights
+                    if t.is_complex():
+                        continue
+                    dist.broadcast(t.data, src=0, group=self.process_group)
 
         self._buckets: List[_Bucket] = []
         self._param_bucket = {}
