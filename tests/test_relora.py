@@ -224,3 +224,29 @@ def test_optimizer_reset_modes(tiny_llama_config):
             reset_optimizer_on_relora=True, optimizer_random_pruning=0.5,
             optimizer_magnitude_pruning=0.0,
         )
+
+
+def test_quantized_save_load_roundtrip(tiny_llama_config, tmp_path):
+    """Quantized wrap -> save_pretrained -> from_pretrained preserves the
+    4-bit payloads and produces identical forward losses."""
+    torch.manual_seed(3)
+    model = LlamaForCausalLM(tiny_llama_config)
+    wrapped = ReLoRaModel(model, r=8, lora_alpha=16, lora_dropout=0.0,
+                          target_modules=["attn", "mlp"], keep_original_weights=True,
+                          quantize="4bit")
+    with torch.no_grad():
+        for n, p in wrapped.named_parameters():
+            if "lora_" in n:
+                p.add_(torch.randn_like(p) * 0.02)
+    wrapped.save_pretrained(tmp_path / "qckpt")
+    reloaded = ReLoRaModel.from_pretrained(str(tmp_path / "qckpt"))
+    lin_a = [m for m in wrapped.modules() if isinstance(m, ReLoRaLinear)][0]
+    lin_b = [m for m in reloaded.modules() if isinstance(m, ReLoRaLinear)][0]
+    assert torch.equal(lin_a.weight.qdata, lin_b.weight.qdata)
+    assert torch.equal(lin_a.weight.absmax, lin_b.weight.absmax)
+    x = torch.randint(0, tiny_llama_config.vocab_size, (1, 16))
+    wrapped.eval(); reloaded.eval()
+    with torch.no_grad():
+        a = wrapped(input_ids=x, labels=x).loss
+        b = reloaded(input_ids=x, labels=x).loss
+    assert torch.allclose(a, b, atol=1e-6)
