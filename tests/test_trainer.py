@@ -274,3 +274,24 @@ def test_training_config_yaml_recipes_parse(tmp_path):
         assert args.total_batch_size == overrides["total_batch_size"]
         assert float(args.lr) == float(overrides["lr"])
         assert args.num_training_steps == int(str(overrides["num_training_steps"]).replace("_", ""))
+
+
+def test_deterministic_training(tmp_path):
+    """Two runs with the same seed produce identical losses and weights
+    (philox dropout, synthetic data, and init are all seed-keyed)."""
+    import torch as _t
+
+    losses = []
+    for run in ("a", "b"):
+        args = run_args(tmp_path / run, steps=6)
+        main(args)
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        sd = _t.load(tmp_path / run / "run" / "model_6" / "pytorch_model.bin",
+                     map_location="cpu", weights_only=True)
+        losses.append(sd)
+    a, b = losses
+    assert set(a) == set(b)
+    for k in a:
+        assert _t.equal(a[k], b[k]), k
