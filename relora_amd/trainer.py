@@ -259,6 +259,13 @@ def main(args):
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
     world_size = int(os.environ.get("WORLD_SIZE", 1))
 
+    # debug mode (SURVEY.md §5 race-detection plan): serialize kernel
+    # launches/copies so HIP faults surface at the offending launch
+    if os.environ.get("RELORA_AMD_DEBUG_SERIALIZE") == "1":
+        os.environ.setdefault("AMD_SERIALIZE_KERNEL", "3")
+        os.environ.setdefault("AMD_SERIALIZE_COPY", "3")
+        logger.warning("RELORA_AMD_DEBUG_SERIALIZE: serializing HIP kernels/copies")
+
     use_gpu = torch.cuda.is_available()
     if use_gpu:
         torch.cuda.set_device(local_rank)

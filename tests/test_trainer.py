@@ -295,3 +295,16 @@ def test_deterministic_training(tmp_path):
     assert set(a) == set(b)
     for k in a:
         assert _t.equal(a[k], b[k]), k
+
+
+def test_profiler_flag(tmp_path, monkeypatch):
+    """--profile true produces torch.profiler traces (reference
+    maybe_make_profiler, torchrun_main.py:322-335)."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    args = run_args(tmp_path, extra=["--profile", "true"])
+    monkeypatch.chdir(tmp_path)  # after parse: config path was resolved in repo
+    args.model_config = os.path.join(repo, "configs", "llama_9m.json")
+    main(args)
+    import glob as _g
+    logs = _g.glob(str(tmp_path / "profiler_logs" / "*" / "*"))
+    assert logs, "no profiler trace written"
