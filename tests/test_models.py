@@ -143,3 +143,71 @@ def test_rope_cache_stays_fp32_after_model_cast():
     rot = pm.gpt_neox.layers[0].attention.rotary_emb
     cos, sin = rot(torch.zeros(1, dtype=torch.bfloat16), seq_len=8)
     assert cos.dtype == torch.float32 and sin.dtype == torch.float32
+
+
+def test_gradient_checkpointing_matches_plain():
+    """Grad checkpointing produces the same loss and gradients."""
+    from relora_amd.models import build_model_from_config, load_model_config
+
+    torch.manual_seed(0)
+    cfg = load_model_config("configs/llama_9m.json")
+    model = build_model_from_config(cfg)
+    x = torch.randint(0, cfg.vocab_size, (2, 32))
+
+    loss_plain = model(input_ids=x, labels=x).loss
+    loss_plain.backward()
+    grads_plain = {n: p.grad.clone() for n, p in model.named_parameters()
+                   if p.grad is not None}
+    model.zero_grad()
+
+    model.gradient_checkpointing_enable()
+    loss_ckpt = model(input_ids=x, labels=x).loss
+    loss_ckpt.backward()
+    assert torch.allclose(loss_plain, loss_ckpt, atol=1e-6)
+    for n, p in model.named_parameters():
+        if p.grad is not None:
+            assert torch.allclose(grads_plain[n], p.grad, atol=1e-5), n
+
+
+def test_kv_cache_incremental_decode_matches_full():
+    """use_cache incremental decode equals the full forward logits
+    (the reference's generation affordance, modeling_llama.py past_key_values)."""
+    from relora_amd.models import build_model_from_config, load_model_config
+
+    torch.manual_seed(1)
+    cfg = load_model_config("configs/llama_9m.json")
+    model = build_model_from_config(cfg).eval()
+    x = torch.randint(0, cfg.vocab_size, (1, 12))
+    with torch.no_grad():
+        full = model(input_ids=x).logits
+        # prefill 8 tokens, then decode 4 one by one
+        out = model(input_ids=x[:, :8], use_cache=True)
+        past = out.past_key_values
+        logits = [out.logits]
+        for t in range(8, 12):
+            step = model(input_ids=x[:, t:t + 1], past_key_values=past, use_cache=True)
+            past = step.past_key_values
+            logits.append(step.logits)
+        inc = torch.cat(logits, dim=1)
+    assert torch.allclose(full, inc, atol=2e-4, rtol=1e-4), (full - inc).abs().max()
+
+
+def test_pythia_rope_scaling_variants():
+    """linear and dynamic-NTK RoPE scaling paths run and differ from base."""
+    from relora_amd.models.pythia import GPTNeoXConfig, GPTNeoXForCausalLM
+
+    torch.manual_seed(2)
+    base_cfg = dict(vocab_size=128, hidden_size=64, num_hidden_layers=1,
+                    num_attention_heads=2, intermediate_size=128,
+                    max_position_embeddings=16, rotary_pct=0.5)
+    x = torch.randint(0, 128, (1, 32))  # beyond max_position -> scaling engages
+    outs = {}
+    for label, scaling in (("base", None),
+                           ("linear", {"type": "linear", "factor": 4.0}),
+                           ("dynamic", {"type": "dynamic", "factor": 4.0})):
+        torch.manual_seed(3)
+        m = GPTNeoXForCausalLM(GPTNeoXConfig(**base_cfg, rope_scaling=scaling)).eval()
+        with torch.no_grad():
+            outs[label] = m(input_ids=x).logits
+    assert not torch.allclose(outs["base"], outs["linear"])
+    assert not torch.allclose(outs["base"], outs["dynamic"])
