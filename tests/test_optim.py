@@ -83,3 +83,36 @@ def test_clip_no_scale_when_below():
     before = params[0].grad.clone()
     clip_grad_norm_(params, 1.0)
     assert torch.equal(params[0].grad, before)
+
+
+def test_delete_old_checkpoints(tmp_path):
+    """--keep_checkpoints GC keeps the N latest model_* dirs
+    (reference training_utils.py:406-418)."""
+    from relora_amd.training_utils import delete_old_checkpoints
+
+    for step in (10, 20, 30, 40):
+        d = tmp_path / f"model_{step}"
+        d.mkdir()
+        (d / "pytorch_model.bin").write_bytes(b"x")
+    (tmp_path / "training_config.yaml").write_text("a: 1")
+    delete_old_checkpoints(str(tmp_path), keep=2)
+    left = sorted(p.name for p in tmp_path.glob("model_*"))
+    assert left == ["model_30", "model_40"]
+    assert (tmp_path / "training_config.yaml").exists()
+    # keep=None is a no-op
+    delete_old_checkpoints(str(tmp_path), keep=None)
+    assert sorted(p.name for p in tmp_path.glob("model_*")) == ["model_30", "model_40"]
+
+
+def test_get_last_training_state(tmp_path):
+    from relora_amd.training_utils import get_last_training_state
+
+    import json as _json
+    for step in (5, 25, 15):
+        d = tmp_path / f"model_{step}"
+        d.mkdir()
+        with open(d / "training_state.json", "w") as f:
+            _json.dump({"update_step": step, "wandb_id": f"id{step}"}, f)
+    state, ckpt = get_last_training_state(str(tmp_path))
+    assert ckpt.endswith("model_25")
+    assert state["update_step"] == 25
