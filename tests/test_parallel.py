@@ -188,3 +188,63 @@ def test_zero1_matches_plain_adamw():
         p.join(timeout=30)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+def _zero_resume_worker(rank, world, port, q):
+    try:
+        _init(rank, world, port)
+        import tempfile
+
+        from relora_amd.parallel import ZeroRedundancyAdamW
+
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+        opt = ZeroRedundancyAdamW(model.parameters(), lr=1e-2)
+        for _ in range(3):
+            loss = model(torch.randn(4, 8)).square().mean()
+            loss.backward()
+            opt.step()
+            opt.zero_grad()
+
+        # save: consolidate to rank 0, broadcast the dict to all (as a file
+        # would be shared in practice)
+        opt.consolidate_state_dict(to=0)
+        payload = [opt.state_dict() if rank == 0 else None]
+        dist.broadcast_object_list(payload, src=0)
+        sd = payload[0]
+
+        # fresh optimizer resumes and must hold identical shard states
+        opt2 = ZeroRedundancyAdamW(model.parameters(), lr=1e-2)
+        opt2.load_state_dict(sd)
+        ok = True
+        for p in opt.shard_params:
+            a = opt.optim.state[p]
+            b = opt2.optim.state[p]
+            for k in ("exp_avg", "exp_avg_sq"):
+                if not torch.allclose(a[k], b[k]):
+                    ok = False
+        # one more identical step on both must produce identical params
+        g = [torch.randn_like(p) for p in model.parameters()]
+        for p, gg in zip(model.parameters(), g):
+            p.grad = gg.clone()
+        opt2.step()
+        q.put((rank, ok, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_zero1_consolidate_resume_roundtrip():
+    port = free_port()
+    q = mp.get_context("spawn").Queue()
+    ps = [mp.get_context("spawn").Process(target=_zero_resume_worker, args=(r, 2, port, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, ok, err in results:
+        assert ok, f"rank {rank}: {err}"
