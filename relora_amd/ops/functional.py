@@ -242,6 +242,10 @@ def flash_attention(q, k, v, causal=True, dropout_p=0.0, scale=None):
 # shapes (16384 x 50304 = 1.6 GB), so default to a single chunk: fewer, larger
 # GEMMs and no fp32 dw accumulation pass. Shrink via env on smaller cards.
 _CE_CHUNK = int(os.environ.get("RELORA_AMD_CE_CHUNK", "16384"))
+# RELORA_AMD_CE_SAVE_LOGITS=1 trades HBM for time: keep the bf16 logits from
+# forward instead of recomputing the [M,V] GEMM in backward (single-chunk
+# mode only; ~1 GB at the flagship shape for ~1.6 ms/step of GEMM).
+_CE_SAVE_LOGITS = os.environ.get("RELORA_AMD_CE_SAVE_LOGITS", "0") == "1"
 
 
 def _row_stats_torch(logits, labels, ignore_index):
@@ -264,9 +268,13 @@ class _FusedCrossEntropy(torch.autograd.Function):
         valid = labels != ignore_index
         n_valid = int(valid.sum().item())
         use_hip = hip.use_hip(hidden)
+        keep_logits = _CE_SAVE_LOGITS and M <= _CE_CHUNK
+        saved_logits = None
         for s in range(0, M, _CE_CHUNK):
             e = min(s + _CE_CHUNK, M)
             logits = hidden[s:e] @ weight.t()
+            if keep_logits:
+                saved_logits = logits
             lab = labels[s:e]
             if use_hip:
                 lse, tgt = hip.ext().ce_row_stats(logits, lab, ignore_index)
@@ -275,14 +283,16 @@ class _FusedCrossEntropy(torch.autograd.Function):
             lse_all[s:e] = lse
             vmask = lab != ignore_index
             loss_sum += torch.where(vmask, lse - tgt, torch.zeros_like(lse)).sum()
-        ctx.save_for_backward(hidden, weight, labels, lse_all)
+        ctx.save_for_backward(hidden, weight, labels, lse_all,
+                              saved_logits if saved_logits is not None
+                              else hidden.new_empty(0))
         ctx.ignore_index = ignore_index
         ctx.n_valid = max(n_valid, 1)
         return loss_sum / max(n_valid, 1)
 
     @staticmethod
     def backward(ctx, grad_out):
-        hidden, weight, labels, lse_all = ctx.saved_tensors
+        hidden, weight, labels, lse_all, saved_logits = ctx.saved_tensors
         ignore_index = ctx.ignore_index
         M, H = hidden.shape
         use_hip = hip.use_hip(hidden)
@@ -293,7 +303,10 @@ class _FusedCrossEntropy(torch.autograd.Function):
         gscale = (grad_out.float() / ctx.n_valid).item() if grad_out.dim() == 0 else None
         for s in range(0, M, _CE_CHUNK):
             e = min(s + _CE_CHUNK, M)
-            logits = hidden[s:e] @ weight.t()
+            if saved_logits.numel() and s == 0 and e == M:
+                logits = saved_logits
+            else:
+                logits = hidden[s:e] @ weight.t()
             lab = labels[s:e]
             lse = lse_all[s:e]
             if use_hip:

@@ -165,3 +165,35 @@ def test_lora_linear_lora_only():
     B = torch.randn(6, 2)
     y = ops.lora_linear(x, None, None, A, B, 0.5, lora_only=True)
     assert torch.allclose(y, (x @ A.t() @ B.t()) * 0.5, atol=1e-6)
+
+
+def test_fused_ce_save_logits_flag(monkeypatch):
+    """RELORA_AMD_CE_SAVE_LOGITS=1 must produce identical loss and grads
+    (it only skips the backward logits recompute)."""
+    import importlib
+
+    import relora_amd.ops.functional as fn
+
+    torch.manual_seed(0)
+    M, H, V = 64, 32, 97
+    hidden = torch.randn(M, H, requires_grad=True)
+    weight = torch.randn(V, H, requires_grad=True)
+    labels = torch.randint(0, V, (M,))
+    labels[::7] = -100
+
+    loss_a = fn.fused_cross_entropy(hidden, weight, labels)
+    loss_a.backward()
+    ga, gw = hidden.grad.clone(), weight.grad.clone()
+    hidden.grad = weight.grad = None
+
+    monkeypatch.setenv("RELORA_AMD_CE_SAVE_LOGITS", "1")
+    importlib.reload(fn)
+    try:
+        loss_b = fn.fused_cross_entropy(hidden, weight, labels)
+        loss_b.backward()
+        assert torch.allclose(loss_a, loss_b, atol=1e-6)
+        assert torch.allclose(ga, hidden.grad, atol=1e-6)
+        assert torch.allclose(gw, weight.grad, atol=1e-6)
+    finally:
+        monkeypatch.delenv("RELORA_AMD_CE_SAVE_LOGITS")
+        importlib.reload(fn)
