@@ -64,8 +64,19 @@ def test_pretokenize_then_train(local_corpus, monkeypatch):
         monkeypatch.delenv(k, raising=False)
     monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
     monkeypatch.setenv("MASTER_PORT", "29751")
+    # model vocab must match the tokenizer vocab (trainer enforces it,
+    # reference torchrun_main.py:481-486)
+    tiny_cfg = {
+        "architectures": ["LlamaForCausalLM"], "model_type": "llama",
+        "hidden_size": 32, "intermediate_size": 64, "num_attention_heads": 4,
+        "num_hidden_layers": 2, "max_position_embeddings": 64,
+        "max_sequence_length": 64, "rms_norm_eps": 1e-6, "hidden_act": "silu",
+        "initializer_range": 0.02, "vocab_size": 63,
+    }
+    cfg_path = tmp_path / "tiny63.json"
+    cfg_path.write_text(json.dumps(tiny_cfg))
     targs = train_args([
-        "--model_config", "configs/llama_9m.json",
+        "--model_config", str(cfg_path),
         "--dataset_path", str(out),
         "--num_training_steps", "3",
         "--batch_size", "2", "--total_batch_size", "2",
