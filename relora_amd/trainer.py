@@ -340,9 +340,12 @@ def main(args):
 
     if dist.is_initialized():
         dist.barrier()
-        run_name = [wandb.run.name] if global_rank == 0 else [""]
-        dist.broadcast_object_list(run_name, src=0)  # C7
-        args.run_name = run_name[0]
+        # C7 + save_dir: both are derived on rank 0 (save_dir from the run
+        # name when not given) — replicate so every rank formats the same
+        # checkpoint paths
+        meta = [wandb.run.name, args.save_dir] if global_rank == 0 else ["", None]
+        dist.broadcast_object_list(meta, src=0)
+        args.run_name, args.save_dir = meta
     else:
         args.run_name = getattr(args, "run_name", "local")
     if args.save_dir is None:
