@@ -713,7 +713,7 @@ def main(args):
                 training_state_checkpoint=training_state_checkpoint,
                 run_config=run_config, save_dir=current_model_directory, dtype=args.dtype,
             )
-            if args.keep_checkpoints is not None:
+            if args.keep_checkpoints is not None and global_rank == 0:
                 training_utils.delete_old_checkpoints(args.save_dir, keep=args.keep_checkpoints)
 
         if update_step % args.eval_every == 0:
