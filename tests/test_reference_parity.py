@@ -159,8 +159,8 @@ def test_scheduler_matches_reference(ref_modules):
     for kwargs in (
         dict(num_training_steps=120, warmup_steps=10, min_lr_ratio=0.1,
              cycle_length=40, restart_warmup_steps=5, adjust_step=0),
-        dict(num_training_steps=200, warmup_steps=20, min_lr_ratio=0.2,
-             cycle_length=40, restart_warmup_steps=10, adjust_step=40),
+        dict(num_training_steps=240, warmup_steps=10, min_lr_ratio=0.2,
+             cycle_length=60, restart_warmup_steps=10, adjust_step=40),
     ):
         p1 = torch.nn.Parameter(torch.zeros(1))
         o1 = torch.optim.SGD([p1], lr=1.0)
