@@ -122,6 +122,14 @@ def test_args_fp16_rejected():
         parse_args(["--synthetic_data", "true", "--batch_size", "2", "--dtype", "float16"])
 
 
+def test_args_fsdp_rejected():
+    """FSDP is hard-disabled, matching the reference (save_model_fsdp raises,
+    torchrun_main.py:227-253)."""
+    with pytest.raises(NotImplementedError):
+        parse_args(["--synthetic_data", "true", "--batch_size", "2",
+                    "--total_batch_size", "2", "--distributed_type", "fsdp"])
+
+
 def test_args_relora_implies_peft():
     args = parse_args(["--synthetic_data", "true", "--batch_size", "2", "--relora", "10"])
     assert args.use_peft is True
