@@ -39,9 +39,13 @@ def make_data_loader(dataset, neox_args):
         rank=rank,
         world_size=world_size,
     )
+    # dedicated generator: DataLoader.__iter__ otherwise draws its _base_seed
+    # from the global RNG, which would offset the global stream by when the
+    # iterator is created and break bit-exact resume
     return torch.utils.data.DataLoader(
         dataset, batch_sampler=batch_sampler,
-        num_workers=neox_args.num_workers, pin_memory=True)
+        num_workers=neox_args.num_workers, pin_memory=True,
+        generator=torch.Generator().manual_seed(int(neox_args.seed) * 100003 + 3))
 
 
 def build_the_dataset(data_prefix, name, data_impl, num_samples, seq_length,

@@ -207,6 +207,21 @@ def save_model(model, *, optimizer, scheduler, training_state_checkpoint, run_co
         logger.info("Consolidating ZeRO optimizer state dict")
         optimizer.consolidate_state_dict()  # C9
 
+    # per-rank RNG streams so resume is bit-exact even with dropout active
+    # (the reference does not restore RNG state across resume; we do better)
+    rng_states = {
+        "torch": torch.get_rng_state(),
+        "numpy": np.random.get_state(),
+        "python": random.getstate(),
+    }
+    if torch.cuda.is_available():
+        rng_states["cuda"] = torch.cuda.get_rng_state()
+    if dist.is_initialized():
+        all_rng_states = [None] * dist.get_world_size()
+        dist.all_gather_object(all_rng_states, rng_states)
+    else:
+        all_rng_states = [rng_states]
+
     if global_rank == 0:
         optimizer_checkpoint = {
             "optimizer": optimizer.state_dict(),
@@ -215,6 +230,7 @@ def save_model(model, *, optimizer, scheduler, training_state_checkpoint, run_co
             "global_step": training_state_checkpoint["global_step"],
             "config": run_config,
             "dtype": dtype,
+            "rng_states_per_rank": all_rng_states,
         }
         torch.save(optimizer_checkpoint, f"{save_dir}/optimizer.pt")
         training_state_checkpoint["wandb_id"] = wandb.run.id if wandb.run else None
@@ -580,6 +596,14 @@ def main(args):
             scheduler.load_state_dict(optimizer_checkpoint["scheduler"])
             update_step = optimizer_checkpoint["update_step"]
             global_step = optimizer_checkpoint["global_step"]
+            _rng = optimizer_checkpoint.get("rng_states_per_rank")
+            if _rng and global_rank < len(_rng):
+                _st = _rng[global_rank]
+                torch.set_rng_state(_st["torch"])
+                np.random.set_state(_st["numpy"])
+                random.setstate(_st["python"])
+                if "cuda" in _st and torch.cuda.is_available():
+                    torch.cuda.set_rng_state(_st["cuda"])
             logger.info(f"Optimizer and scheduler restored from {args.resume_from}")
         _tc_path = os.path.join(args.resume_from, "training_config.yaml")
         if os.path.exists(_tc_path):
@@ -589,6 +613,15 @@ def main(args):
                 raise RuntimeError("Cannot resume from a checkpoint with a different batch size.")
 
     # ---- dataloaders -------------------------------------------------------
+    # Every DataLoader gets a dedicated seeded generator: DataLoader.__iter__
+    # draws one int64 from the GLOBAL RNG for its _base_seed, so without this
+    # the global stream would be offset by where/when iterators are created —
+    # which breaks bit-exact resume (a fresh run creates the train iterator at
+    # step 0, a resumed run at the resume step; everything downstream of the
+    # merge re-init would then diverge).
+    def _loader_gen(salt):
+        return torch.Generator().manual_seed(args.seed * 100003 + salt)
+
     if args.dataset_path is not None:
         import datasets.distributed
         train_dataset = datasets.distributed.split_dataset_by_node(
@@ -599,10 +632,11 @@ def main(args):
         train_loader = SkipDataLoader(
             train_dataset, batch_size=args.batch_size, collate_fn=default_data_collator,
             skip_batches=_skip_batches, num_workers=args.workers,
+            generator=_loader_gen(1),
         )
         eval_loader = torch.utils.data.DataLoader(
             eval_dataset, batch_size=args.batch_size, collate_fn=default_data_collator,
-            num_workers=args.workers,
+            num_workers=args.workers, generator=_loader_gen(2),
         )
     elif args.synthetic_data:
         # deterministic contiguous shard per rank
@@ -613,10 +647,11 @@ def main(args):
         train_loader = SkipDataLoader(
             train_dataset, batch_size=args.batch_size, collate_fn=default_data_collator,
             skip_batches=_skip, num_workers=min(args.workers, 2), sampler=sampler,
+            generator=_loader_gen(1),
         )
         eval_loader = torch.utils.data.DataLoader(
             eval_dataset, batch_size=args.batch_size, collate_fn=default_data_collator,
-            num_workers=0,
+            num_workers=0, generator=_loader_gen(2),
         )
     else:
         assert train_loader is not None and eval_loader is not None

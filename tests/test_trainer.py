@@ -70,6 +70,44 @@ def test_trainer_autoresume(tmp_path):
     assert state["tokens_seen"] > 0
 
 
+def test_resume_is_bit_exact(tmp_path):
+    """Interrupted-then-resumed training produces exactly the weights and
+    optimizer state of an uninterrupted run: optimizer/scheduler restore,
+    dataloader fast-forward, and per-rank RNG stream restore (dropout is
+    active through lora_dropout) all have to line up.  The reference does
+    not restore RNG states, so it cannot make this guarantee."""
+    import torch.distributed as dist
+
+    # uninterrupted: 6 steps, checkpoint at 3 and 6
+    main(run_args(tmp_path / "full", extra=["--save_every", "3"], steps=6))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    # interrupted: stop at 3 ...
+    main(run_args(tmp_path / "half", extra=["--save_every", "3"], steps=3))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    # ... resume to 6
+    main(run_args(tmp_path / "half", extra=["--save_every", "3",
+                                            "--autoresume", "true"], steps=6))
+
+    a = torch.load(tmp_path / "full" / "run" / "model_6" / "pytorch_model.bin",
+                   map_location="cpu", weights_only=True)
+    b = torch.load(tmp_path / "half" / "run" / "model_6" / "pytorch_model.bin",
+                   map_location="cpu", weights_only=True)
+    assert set(a) == set(b)
+    for k in a:
+        assert torch.equal(a[k], b[k]), k
+    oa = torch.load(tmp_path / "full" / "run" / "model_6" / "optimizer.pt",
+                    map_location="cpu", weights_only=False)
+    ob = torch.load(tmp_path / "half" / "run" / "model_6" / "optimizer.pt",
+                    map_location="cpu", weights_only=False)
+    sa, sb = oa["optimizer"]["state"], ob["optimizer"]["state"]
+    assert set(map(str, sa)) == set(map(str, sb))
+    for k in sa:
+        for key in ("exp_avg", "exp_avg_sq"):
+            assert torch.equal(sa[k][key], sb[k][key]), (k, key)
+
+
 def test_trainer_zero_optimizer(tmp_path):
     args = run_args(tmp_path, extra=["--optimizer", "adam_zero"])
     main(args)

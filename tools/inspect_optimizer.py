@@ -36,7 +36,12 @@ def main(argv=None):
     path = args.path
     if os.path.isdir(path):
         path = os.path.join(path, "optimizer.pt")
-    ckpt = torch.load(path, map_location="cpu", weights_only=True)
+    try:
+        ckpt = torch.load(path, map_location="cpu", weights_only=True)
+    except Exception:
+        # the checkpoint carries saved RNG states (numpy tuples), which the
+        # weights_only unpickler rejects; these are our own local checkpoints
+        ckpt = torch.load(path, map_location="cpu", weights_only=False)
     opt_sd = ckpt.get("optimizer", ckpt)
     if "update_step" in ckpt:
         print(f"update_step {ckpt['update_step']}  global_step "
