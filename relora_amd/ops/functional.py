@@ -349,6 +349,15 @@ def fused_cross_entropy(hidden, weight, labels, ignore_index=-100):
 _lora_seed_counter = [0]
 
 
+def get_dropout_rng_state():
+    """Philox dropout stream position (for bit-exact checkpoint resume)."""
+    return _lora_seed_counter[0]
+
+
+def set_dropout_rng_state(counter):
+    _lora_seed_counter[0] = int(counter)
+
+
 def _next_dropout_seed():
     _lora_seed_counter[0] += 1
     # mix with the torch seed so runs differ when the user reseeds, while

@@ -209,10 +209,12 @@ def save_model(model, *, optimizer, scheduler, training_state_checkpoint, run_co
 
     # per-rank RNG streams so resume is bit-exact even with dropout active
     # (the reference does not restore RNG state across resume; we do better)
+    from relora_amd.ops import functional as _ops_functional
     rng_states = {
         "torch": torch.get_rng_state(),
         "numpy": np.random.get_state(),
         "python": random.getstate(),
+        "philox_dropout": _ops_functional.get_dropout_rng_state(),
     }
     if torch.cuda.is_available():
         rng_states["cuda"] = torch.cuda.get_rng_state()
@@ -604,6 +606,9 @@ def main(args):
                 random.setstate(_st["python"])
                 if "cuda" in _st and torch.cuda.is_available():
                     torch.cuda.set_rng_state(_st["cuda"])
+                if "philox_dropout" in _st:
+                    from relora_amd.ops import functional as _ops_functional
+                    _ops_functional.set_dropout_rng_state(_st["philox_dropout"])
             logger.info(f"Optimizer and scheduler restored from {args.resume_from}")
         _tc_path = os.path.join(args.resume_from, "training_config.yaml")
         if os.path.exists(_tc_path):
