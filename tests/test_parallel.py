@@ -248,3 +248,79 @@ def test_zero1_consolidate_resume_roundtrip():
         p.join(timeout=60)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+def _world4_worker(rank, world, port, q):
+    """DDP averaging + ZeRO-1 at world_size=4 with UNEVEN shards (5 params
+    over 4 ranks) — the multi-GPU shape the driver's 8-GPU scaling run
+    exercises, minus the hardware."""
+    try:
+        _init(rank, world, port)
+        from relora_amd.parallel import DistributedModel, ZeroRedundancyAdamW
+        from relora_amd.ops.optim import AdamW
+
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(8, 12), torch.nn.ReLU(),
+            torch.nn.Linear(12, 6, bias=True), torch.nn.Linear(6, 4, bias=False),
+        )
+        wrapped = DistributedModel(model, bucket_cap_mb=0.0001)
+        params = [p for p in model.parameters() if p.requires_grad]
+        opt = ZeroRedundancyAdamW(params, lr=1e-2, betas=(0.9, 0.999),
+                                  weight_decay=0.01)
+
+        ref = torch.nn.Sequential(
+            torch.nn.Linear(8, 12), torch.nn.ReLU(),
+            torch.nn.Linear(12, 6, bias=True), torch.nn.Linear(6, 4, bias=False),
+        )
+        ref.load_state_dict(model.state_dict())
+        ref_opt = AdamW(list(ref.parameters()), lr=1e-2, betas=(0.9, 0.999),
+                        weight_decay=0.01)
+
+        torch.manual_seed(7)
+        for it in range(3):
+            x = torch.randn(world * 2, 8)
+            y = torch.randn(world * 2, 4)
+            xb, yb = x[rank * 2: rank * 2 + 2], y[rank * 2: rank * 2 + 2]
+            wrapped.set_gradient_sync(True)
+            torch.nn.functional.mse_loss(wrapped(xb), yb).backward()
+            wrapped.finish_gradient_sync()
+            opt.step()
+            wrapped.zero_grad_buffers()
+
+            # oracle: mean over the 4 per-rank losses on one process
+            ref_loss = sum(
+                torch.nn.functional.mse_loss(ref(x[r * 2: r * 2 + 2]),
+                                             y[r * 2: r * 2 + 2])
+                for r in range(world)) / world
+            ref_loss.backward()
+            ref_opt.step()
+            ref_opt.zero_grad()
+
+        ok = all(torch.allclose(p, rp, atol=1e-5)
+                 for p, rp in zip(model.parameters(), ref.parameters()))
+        opt.consolidate_state_dict()
+        if rank == 0:
+            sd = opt.state_dict()
+            ok = ok and len(sd["state"]) == len(params)
+        q.put((rank, ok, None))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ddp_zero1_world4():
+    world, port = 4, free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_world4_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok, err in results:
+        assert ok, f"rank {rank}: {err}"
