@@ -294,3 +294,44 @@ def test_optimizer_reset_matches_reference(ref_modules):
         s1, s2 = o1.state[p1], o2.state[p2]
         for k in ("exp_avg", "exp_avg_sq"):
             assert torch.equal(s1[k], s2[k])
+
+
+def test_tokenize_and_chunk_matches_reference(ref_modules, tmp_path):
+    """HF data path: identical tokenizer + corpus -> identical chunked token
+    streams (reference dataloader.py:57-124)."""
+    del ref_modules
+    sys.path.insert(0, REF)
+    try:
+        from peft_pretraining import dataloader as ref_dl
+    finally:
+        sys.path.remove(REF)
+    import datasets as hfds
+    from tokenizers import Tokenizer, models, pre_tokenizers
+    from transformers import PreTrainedTokenizerFast
+
+    from relora_amd.data.dataloader import tokenize_and_chunk as our_tac
+
+    words = [f"w{i}" for i in range(40)]
+    vocab = {"[PAD]": 0, "[UNK]": 1, "</s>": 2}
+    for w in words:
+        vocab[w] = len(vocab)
+    tok = Tokenizer(models.WordLevel(vocab=vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    fast = PreTrainedTokenizerFast(tokenizer_object=tok, pad_token="[PAD]",
+                                   unk_token="[UNK]", eos_token="</s>")
+
+    import numpy as np
+    rng = np.random.RandomState(3)
+    texts = [" ".join(rng.choice(words, size=rng.randint(3, 30)))
+             for _ in range(60)]
+    # the reference asserts on a "train" split, i.e. expects a DatasetDict
+    ds_ref = hfds.DatasetDict({"train": hfds.Dataset.from_dict({"text": list(texts)})})
+    ds_our = hfds.DatasetDict({"train": hfds.Dataset.from_dict({"text": list(texts)})})
+
+    ref_out = ref_dl.tokenize_and_chunk(fast, ds_ref, "text", 16, num_cpu=1)["train"]
+    our_out = our_tac(fast, ds_our, "text", 16, num_cpu=1)["train"]
+
+    assert len(ref_out) == len(our_out)
+    assert set(ref_out.column_names) == set(our_out.column_names)
+    for i in range(len(ref_out)):
+        assert ref_out[i]["input_ids"] == our_out[i]["input_ids"], i
