@@ -18,13 +18,29 @@ def save_pretrained_compat(model, save_dir):
     torch.save(model.state_dict(), os.path.join(save_dir, "pytorch_model.bin"))
 
 
+# Buffer keys the reference persists that our models recompute (fp32 RoPE
+# caches are non-persistent here; the attention bias mask is implicit in the
+# flash kernel).  Dropped on load so reference-written checkpoints load
+# strict (reference modeling_llama.py rotary_emb.inv_freq,
+# modeling_pythia.py attention.bias/masked_bias + cos/sin caches).
+_LEGACY_BUFFER_KEYS = ("rotary_emb.inv_freq", "attention.bias",
+                       "attention.masked_bias", "cos_cached", "sin_cached")
+
+
+def _drop_legacy_buffers(sd):
+    return {k: v for k, v in sd.items()
+            if not any(k.endswith(s) or s in k for s in _LEGACY_BUFFER_KEYS)}
+
+
 def load_state_dict_compat(path):
-    """Load a state dict from `pytorch_model.bin` or `model.safetensors`."""
+    """Load a state dict from `pytorch_model.bin` or `model.safetensors`,
+    dropping the reference's persistent rotary/mask buffers."""
     bin_path = os.path.join(path, "pytorch_model.bin")
     if os.path.exists(bin_path):
-        return torch.load(bin_path, map_location="cpu", weights_only=True)
+        return _drop_legacy_buffers(
+            torch.load(bin_path, map_location="cpu", weights_only=True))
     st_path = os.path.join(path, "model.safetensors")
     if os.path.exists(st_path):
         from safetensors.torch import load_file
-        return load_file(st_path)
+        return _drop_legacy_buffers(load_file(st_path))
     raise FileNotFoundError(f"No model weights found under {path}")

@@ -335,3 +335,103 @@ def test_tokenize_and_chunk_matches_reference(ref_modules, tmp_path):
     assert set(ref_out.column_names) == set(our_out.column_names)
     for i in range(len(ref_out)):
         assert ref_out[i]["input_ids"] == our_out[i]["input_ids"], i
+
+
+def test_checkpoint_cross_compatibility(ref_modules, tmp_path):
+    """A user switching frameworks can load checkpoints across: a checkpoint
+    written by the REFERENCE's ReLoRaModel.save_pretrained loads through OUR
+    ReLoRaModel.from_pretrained with identical forward, and vice versa."""
+    ref_llama, ref_relora = ref_modules
+    hf_cfg, our_cfg = _small_cfg_pair(ref_llama)
+
+    from relora_amd.models.llama import LlamaForCausalLM as OurLlama
+    from relora_amd.relora import ReLoRaModel as OurReLoRa
+
+    torch.manual_seed(11)
+    base = ref_llama.LlamaForCausalLM(hf_cfg)
+    ref_wrapped = ref_relora.ReLoRaModel(
+        base, r=8, lora_alpha=32, lora_dropout=0.0,
+        target_modules=["attn", "attention", "mlp"],
+        keep_original_weights=True)
+    with torch.no_grad():
+        for n, p in ref_wrapped.named_parameters():
+            if "lora_" in n:
+                p.copy_(torch.randn_like(p) * 0.03)
+
+    # reference -> ours
+    ref_dir = tmp_path / "ref_ckpt"
+    ref_wrapped.save_pretrained(str(ref_dir))
+    # the reference's save relies on HF save_pretrained, whose modern layout
+    # writes model.safetensors; our loader accepts both, but make sure the
+    # reference config is readable by our LlamaConfig loader
+    ours_loaded = OurReLoRa.from_pretrained(str(ref_dir))
+
+    x = torch.randint(0, 128, (2, 16))
+    ref_wrapped.eval(); ours_loaded.eval()
+    with torch.no_grad():
+        rl = ref_wrapped(input_ids=x, labels=x).loss
+        ol = ours_loaded(input_ids=x, labels=x).loss
+    assert torch.allclose(rl, ol, atol=1e-6), (rl, ol)
+
+    # ours -> reference
+    torch.manual_seed(12)
+    ours_base = OurLlama(our_cfg)
+    our_wrapped = OurReLoRa(ours_base, r=8, lora_alpha=32, lora_dropout=0.0,
+                            target_modules=["attn", "attention", "mlp"],
+                            keep_original_weights=True)
+    with torch.no_grad():
+        for n, p in our_wrapped.named_parameters():
+            if "lora_" in n:
+                p.copy_(torch.randn_like(p) * 0.03)
+    our_dir = tmp_path / "our_ckpt"
+    our_wrapped.save_pretrained(str(our_dir))
+    ref_loaded = ref_relora.ReLoRaModel.from_pretrained(str(our_dir))
+    ref_loaded.eval(); our_wrapped.eval()
+    with torch.no_grad():
+        rl2 = ref_loaded(input_ids=x, labels=x).loss
+        ol2 = our_wrapped(input_ids=x, labels=x).loss
+    assert torch.allclose(rl2, ol2, atol=1e-6), (rl2, ol2)
+
+
+def test_megatron_index_builders_match_reference(ref_modules):
+    """GPT2Dataset index maps: our builders produce the exact arrays the
+    reference's python fallback produces (dataset.py:275-330), and the same
+    shuffle/doc orderings under the same seed."""
+    del ref_modules
+    sys.path.insert(0, REF)
+    try:
+        from peft_pretraining.megatron_dataset import dataset as ref_ds
+    finally:
+        sys.path.remove(REF)
+    import numpy as np
+
+    from relora_amd.data import gpt2_dataset as our_ds
+
+    rng = np.random.RandomState(0)
+    for trial in range(5):
+        n_docs = int(rng.randint(2, 50))
+        sizes = rng.randint(1, 60, size=n_docs).astype(np.int32)
+        seq_length = int(rng.randint(2, 33))
+        num_epochs = int(rng.randint(1, 4))
+        seed = int(rng.randint(0, 10000))
+
+        np_rng_ref = np.random.RandomState(seed)
+        np_rng_our = np.random.RandomState(seed)
+        documents = np.arange(n_docs, dtype=np.int32)
+        doc_idx_ref = ref_ds._build_doc_idx(documents, num_epochs, np_rng_ref)
+        doc_idx_our = our_ds._build_doc_idx(documents, num_epochs, np_rng_our)
+        np.testing.assert_array_equal(doc_idx_ref, doc_idx_our)
+
+        tokens_per_epoch = int(sizes.sum())
+        ref_sample = ref_ds._build_sample_idx(sizes, doc_idx_ref, seq_length,
+                                              num_epochs, tokens_per_epoch)
+        our_sample = our_ds.build_sample_idx_py(sizes, doc_idx_our, seq_length,
+                                                num_epochs, tokens_per_epoch)
+        np.testing.assert_array_equal(ref_sample, our_sample)
+
+        num_samples = ref_sample.shape[0] - 1
+        ref_shuffle = ref_ds._build_shuffle_idx(num_samples,
+                                                np.random.RandomState(seed + 1))
+        our_shuffle = our_ds._build_shuffle_idx(num_samples,
+                                                np.random.RandomState(seed + 1))
+        np.testing.assert_array_equal(ref_shuffle, our_shuffle)
