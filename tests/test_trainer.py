@@ -354,3 +354,16 @@ def test_profiler_flag(tmp_path, monkeypatch):
     import glob as _g
     logs = _g.glob(str(tmp_path / "profiler_logs" / "*" / "*"))
     assert logs, "no profiler trace written"
+
+
+def test_regime_quantized_relora(tmp_path):
+    """--quantize 4bit end-to-end on CPU: frozen weights stored NF4, training
+    steps run, merge requantizes, checkpoint saves (reference regime:
+    relora.py 4-bit bnb path; ours is the HIP NF4 kernel / python oracle)."""
+    args = run_args(tmp_path, extra=["--quantize", "4bit"])
+    main(args)
+    ckpt = tmp_path / "run" / "model_6"
+    assert ckpt.exists()
+    state = json.load(open(ckpt / "training_state.json"))
+    assert state["update_step"] == 6
+    assert state["n_lora_restarts"] >= 1
