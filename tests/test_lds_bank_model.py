@@ -75,6 +75,36 @@ def test_lora_tr64_patterns_cap_at_2way():
         (xt_ways, pt_ways, rd_ways)
 
 
+def test_v2_layout_fully_conflict_free_hd64():
+    """The round-2 candidate layout (Q_V2 table) removes the residual
+    2-way read class of the current rotation: conflict-free on BOTH
+    writes and reads at hd64."""
+    from lds_bank_model import attn_t_elem_v2
+    wok, w_ways = verify(attn_tile_write_t_instructions(64, attn_t_elem_v2),
+                         "write_b16")
+    rok, r_ways = verify(attn_ldsT_frag_instructions(64, attn_t_elem_v2),
+                         "read_b128")
+    assert wok and w_ways == 1, w_ways
+    assert rok and r_ways == 1, r_ways
+
+
+def test_v2_layout_hd128_reads_conflict_free_writes_at_floor():
+    from lds_bank_model import attn_t_elem_v2
+    _, w_ways = verify(attn_tile_write_t_instructions(128, attn_t_elem_v2),
+                       "write_b16")
+    rok, r_ways = verify(attn_ldsT_frag_instructions(128, attn_t_elem_v2),
+                         "read_b128")
+    assert w_ways == 2, w_ways  # provable floor of the stride-64 family
+    assert rok and r_ways == 1, r_ways
+
+
+def test_v2_layout_bijective():
+    from lds_bank_model import attn_t_elem_v2
+    for hd in (64, 128):
+        seen = {attn_t_elem_v2(c, kv) for c in range(hd) for kv in range(64)}
+        assert len(seen) == hd * 64
+
+
 def test_rotation_preserves_layout_bijectivity():
     """Sanity: the rotated mapping is a bijection on the 64x64 tile (no
     element aliasing) for both layout families."""

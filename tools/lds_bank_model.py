@@ -105,6 +105,35 @@ def attn_t_elem_naive(c, kv):
     return c * TILE + kv
 
 
+# ---------------------------------------------------------------------------
+# v2 candidate layout (round-2): found by min-conflicts search over the
+# reduced pattern Q[col][g] with per-16-block shift r(t)=2t (the shift turns
+# the cross-block write constraint into a local odd-difference condition).
+# Fully conflict-free at hd64 (writes AND reads); hd128 reads conflict-free,
+# writes at the provable 2-way floor of the stride-64 family (bank of a b16
+# write is (f*4 + (kv&7)/2) mod 32 — the c*64 stride contributes 0 mod 32,
+# so 16 channel-blocks per instruction must share 8 rotation values).
+# Verified by tests/test_lds_bank_model.py::test_v2_layout_*.
+# ---------------------------------------------------------------------------
+
+Q_V2 = [
+    [4, 2, 0, 3, 6, 1, 7, 5], [6, 0, 3, 4, 1, 5, 2, 7],
+    [0, 6, 5, 1, 7, 3, 4, 2], [0, 6, 1, 7, 5, 2, 3, 4],
+    [1, 5, 2, 7, 6, 4, 0, 3], [2, 5, 3, 0, 1, 4, 6, 7],
+    [4, 6, 5, 3, 0, 1, 7, 2], [1, 4, 2, 6, 7, 3, 5, 0],
+    [3, 1, 7, 4, 5, 2, 6, 0], [7, 3, 0, 5, 6, 2, 1, 4],
+    [5, 7, 4, 6, 2, 0, 3, 1], [3, 1, 6, 4, 0, 7, 2, 5],
+    [2, 0, 1, 6, 3, 7, 5, 4], [7, 4, 2, 5, 6, 3, 1, 0],
+    [3, 7, 2, 0, 5, 4, 6, 1], [2, 5, 7, 1, 0, 4, 6, 3],
+]
+
+
+def attn_t_elem_v2(c, kv):
+    """v2 transposed-tile layout: elem(c,kv) ->
+    c*64 + ((Q_V2[c&15][kv>>3] + 2*(c>>4)) & 7)*8 + (kv&7)."""
+    return c * 64 + ((Q_V2[c & 15][kv >> 3] + 2 * (c >> 4)) & 7) * 8 + (kv & 7)
+
+
 def rot8(row, c64):
     """lora_gemm.hip:37."""
     return ((((c64 >> 3) + (row >> 3) + (row & 7)) & 7) << 3) + (c64 & 7)
