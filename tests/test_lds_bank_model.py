@@ -105,6 +105,38 @@ def test_v2_layout_bijective():
         assert len(seen) == hd * 64
 
 
+def test_v2_layout_lora_orientation():
+    """Q_V2 on the lora tr64 orientation: reads conflict-free, writes at
+    the 2-way floor, bijective over the 128x64 tile."""
+    from lds_bank_model import tr64_v2
+
+    def stage_writes():
+        for it in range(2):
+            for w in range(8):
+                for j in range(8):
+                    addrs = {}
+                    for lane in range(64):
+                        t = w * 64 + lane + it * 512
+                        if t >= 64 * 16:
+                            continue
+                        addrs[lane] = 2 * tr64_v2((t % 16) * 8 + j, t // 16)
+                    if addrs:
+                        yield addrs
+
+    def frag_reads():
+        for row0 in range(0, 128, 16):
+            for k0 in (0, 32):
+                yield {lane: 2 * tr64_v2(row0 + (lane & 15),
+                                         k0 + (lane >> 4) * 8)
+                       for lane in range(64)}
+
+    _, w_ways = verify(stage_writes(), "write_b16")
+    ok, r_ways = verify(frag_reads(), "read_b128")
+    assert w_ways == 2, w_ways
+    assert ok and r_ways == 1, r_ways
+    assert len({tr64_v2(r, c) for r in range(128) for c in range(64)}) == 128 * 64
+
+
 def test_rotation_preserves_layout_bijectivity():
     """Sanity: the rotated mapping is a bijection on the 64x64 tile (no
     element aliasing) for both layout families."""

@@ -134,6 +134,15 @@ def attn_t_elem_v2(c, kv):
     return c * 64 + ((Q_V2[c & 15][kv >> 3] + 2 * (c >> 4)) & 7) * 8 + (kv & 7)
 
 
+def tr64_v2(row, c):
+    """Q_V2 applied to the lora_gemm orientation (rows up to 128, 64 cols):
+    reads become conflict-free (tr64 is 2-way today); writes stay at the
+    2-way floor.  The odd-difference property Q_V2[8+j]-Q_V2[j] guarantees
+    the 16-row-block write instructions hit each rotation value exactly
+    twice."""
+    return row * 64 + ((Q_V2[row & 15][c >> 3] + 2 * (row >> 4)) & 7) * 8 + (c & 7)
+
+
 def rot8(row, c64):
     """lora_gemm.hip:37."""
     return ((((c64 >> 3) + (row >> 3) + (row & 7)) & 7) << 3) + (c64 & 7)
