@@ -110,14 +110,34 @@ DEV_INLINE void tile_write_rows(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
 }
 
 // Transposed tiles use a rotated layout: element (c, kv) lives at
-// c*64 + (((kv>>3) + (c>>3) + (c&7)) & 7)*8 + (kv&7).  Found by exhaustive
-// search: both the 8-scalar-per-thread transpose WRITES and the bf16x8
-// fragment READS are bank-conflict-free (the naive dst[c][kv] layout put
-// every column write of one instruction on one bank - 8-way, 16% of wave
-// cycles in the dkdv kernel).  No row padding needed: stride is exactly 64.
+// c*64 + t_rot(kv>>3, c)*8 + (kv&7).  The rotation keeps the 8-scalar
+// transpose writes conflict-free (hd64) and caps reads at 2-way — verified
+// by the in-tree bank model (tools/lds_bank_model.py); the naive dst[c][kv]
+// layout put every column write of one instruction on one bank (8-16-way,
+// 16% of dkdv wave cycles).  No row padding needed: stride is exactly 64.
+//
+// RELORA_AMD_ROT_V2: table-driven rotation (Q_V2 from the bank-model
+// search) — fully conflict-free reads AND writes at hd64, conflict-free
+// reads at hd128.  Numerics-neutral (same bijection used by writer and
+// reader); compile with -DRELORA_AMD_ROT_V2=1 to A/B.
+#ifdef RELORA_AMD_ROT_V2
+__device__ constexpr unsigned char Q_V2[16][8] = {
+    {4, 2, 0, 3, 6, 1, 7, 5}, {6, 0, 3, 4, 1, 5, 2, 7},
+    {0, 6, 5, 1, 7, 3, 4, 2}, {0, 6, 1, 7, 5, 2, 3, 4},
+    {1, 5, 2, 7, 6, 4, 0, 3}, {2, 5, 3, 0, 1, 4, 6, 7},
+    {4, 6, 5, 3, 0, 1, 7, 2}, {1, 4, 2, 6, 7, 3, 5, 0},
+    {3, 1, 7, 4, 5, 2, 6, 0}, {7, 3, 0, 5, 6, 2, 1, 4},
+    {5, 7, 4, 6, 2, 0, 3, 1}, {3, 1, 6, 4, 0, 7, 2, 5},
+    {2, 0, 1, 6, 3, 7, 5, 4}, {7, 4, 2, 5, 6, 3, 1, 0},
+    {3, 7, 2, 0, 5, 4, 6, 1}, {2, 5, 7, 1, 0, 4, 6, 3}};
+DEV_INLINE int t_rot(int kv_grp, int c) {
+  return (Q_V2[c & 15][kv_grp] + 2 * (c >> 4)) & 7;
+}
+#else
 DEV_INLINE int t_rot(int kv_grp, int c) {
   return ((kv_grp + (c >> 3) + (c & 7)) & 7);
 }
+#endif
 
 // read a transposed-tile B fragment: channel row c, 8 kv at kv0 (mult of 8)
 DEV_INLINE bf16x8 ldsT_frag(const __bf16* tile, int c, int kv0) {

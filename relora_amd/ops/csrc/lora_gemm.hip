@@ -29,14 +29,30 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define LPAD 8  // bf16 elements of LDS row padding (one 16B slot)
 
 // Rotated layout for transposed LDS tiles with 64-element rows: element
-// (row, c) of an [R][64] image lives at row*64 + rot*8 + (c&7), with
-// rot = ((c>>3) + (row>>3) + (row&7)) & 7.  Exhaustive-search layout (see
-// attention.hip): both 8-scalar transpose writes (8 consecutive rows OR 8
-// consecutive c) and bf16x8 fragment reads are bank-conflict-free, and no
-// row padding is needed.
+// (row, c) of an [R][64] image lives at row*64 + rot8(row, c), with the
+// rotation verified by the in-tree bank model (tools/lds_bank_model.py):
+// writes and reads cap at 2-way with zero padding (the naive layout was
+// 8-way — 44% of skinny_grad wave cycles).  RELORA_AMD_ROT_V2 switches to
+// the Q_V2 table (conflict-free reads, writes stay at the 2-way floor);
+// numerics-neutral — writer and reader share the one bijection.
+#ifdef RELORA_AMD_ROT_V2
+__device__ constexpr unsigned char Q_V2_L[16][8] = {
+    {4, 2, 0, 3, 6, 1, 7, 5}, {6, 0, 3, 4, 1, 5, 2, 7},
+    {0, 6, 5, 1, 7, 3, 4, 2}, {0, 6, 1, 7, 5, 2, 3, 4},
+    {1, 5, 2, 7, 6, 4, 0, 3}, {2, 5, 3, 0, 1, 4, 6, 7},
+    {4, 6, 5, 3, 0, 1, 7, 2}, {1, 4, 2, 6, 7, 3, 5, 0},
+    {3, 1, 7, 4, 5, 2, 6, 0}, {7, 3, 0, 5, 6, 2, 1, 4},
+    {5, 7, 4, 6, 2, 0, 3, 1}, {3, 1, 6, 4, 0, 7, 2, 5},
+    {2, 0, 1, 6, 3, 7, 5, 4}, {7, 4, 2, 5, 6, 3, 1, 0},
+    {3, 7, 2, 0, 5, 4, 6, 1}, {2, 5, 7, 1, 0, 4, 6, 3}};
+DEV_INLINE int rot8(int row, int c64) {
+  return (((Q_V2_L[row & 15][c64 >> 3] + 2 * (row >> 4)) & 7) << 3) + (c64 & 7);
+}
+#else
 DEV_INLINE int rot8(int row, int c64) {
   return ((((c64 >> 3) + (row >> 3) + (row & 7)) & 7) << 3) + (c64 & 7);
 }
+#endif
 DEV_INLINE int tr64(int row, int c) { return row * 64 + rot8(row, c); }
 
 // ---------------------------------------------------------------------------

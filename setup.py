@@ -41,7 +41,12 @@ ext = CUDAExtension(
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
-        "nvcc": ["-O3", "-std=c++17"],
+        # RELORA_AMD_ROT_V2=1 at build time compiles the Q_V2 table-driven
+        # LDS rotation (bank-model-verified conflict-free variant) for A/B;
+        # default is the shipped t_rot/rot8 rotation.
+        "nvcc": ["-O3", "-std=c++17"]
+        + (["-DRELORA_AMD_ROT_V2=1"]
+           if os.environ.get("RELORA_AMD_ROT_V2") == "1" else []),
     },
 )
 
