@@ -14,12 +14,14 @@
 // block barrier.  Two __syncthreads per tile (the v1 kernel paid six per
 // 64 kv rows and ran at ~70 TF).
 //
-// Row-major tiles are padded by 8 bf16 (16 B): row stride 144 B = 36
-// dwords, so 16-lane ds_read_b128 groups of consecutive rows land on
-// disjoint 4-dword bank windows (36*r mod 64 covers all banks) — reads are
-// conflict-free.  Transposed tiles (V^T, K^T, Q^T, dO^T) use the rotated
-// layout of t_rot() below: conflict-free on both the transpose writes and
-// the fragment reads.
+// Row-major tiles are padded by 8 bf16 (16 B) by default: staging writes
+// are conflict-free and fragment reads 2-way (bank model,
+// tools/lds_bank_model.py).  Transposed tiles (V^T, K^T, Q^T, dO^T) use
+// the rotated layout of t_rot() below: transpose writes conflict-free at
+// hd64, everything else capped at 2-way.  Building with
+// -DRELORA_AMD_ROT_V2=1 switches to the bank-model-solved variants
+// (Q_V2 rotation + zero-pad XOR-swizzled row-major tiles) that are
+// conflict-free on every pattern the model covers.
 //
 // Online softmax runs fully in registers on the MFMA C-layout (row r of a
 // 16x16 tile lives in the 16 consecutive lanes with l>>4 == r>>2 at
@@ -39,8 +41,34 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// Row-major LDS tiles: default = 8-element row padding (halves the classic
+// same-slot conflict; K/P fragment reads stay 2-way per the bank model).
+// RELORA_AMD_ROT_V2 = zero padding + per-row XOR swizzle on the element
+// index — bank-model-verified conflict-free on every K/P write and read at
+// hd64 and hd128, and saves the padding LDS.  The swizzle depth follows the
+// row stride: 256-byte rows need (row&15), 128-byte rows (row&7).
+#ifdef RELORA_AMD_ROT_V2
+#define LPAD 0
+#else
 #define LPAD 8     // bf16 elements of row padding in LDS tiles (16B)
+#endif
 #define TILE 64    // staged tile rows (kv rows fwd/dq, q rows dkdv)
+
+DEV_INLINE int row_swz(int row, int ldst) {
+#ifdef RELORA_AMD_ROT_V2
+  return ((ldst & 127) == 0 ? (row & 15) : (row & 7)) << 3;
+#else
+  (void)row; (void)ldst;
+  return 0;
+#endif
+}
+
+// swizzled element index into a row-major [R][ldst] LDS tile — EVERY
+// producer and consumer of these tiles must address through this (or
+// lds_frag/tile_write_rows, which do)
+DEV_INLINE int lds_rm_idx(int row, int col, int ldst) {
+  return (row * ldst + col) ^ row_swz(row, ldst);
+}
 
 DEV_INLINE float bf_to_f(__bf16 x) { return (float)x; }
 
@@ -59,7 +87,7 @@ DEV_INLINE float rowgroup_sum(float x) {
 // load an A/B fragment from an LDS tile: lane reads row `row`, 8 elements
 // at column k0. Caller guarantees 16B alignment (ldst multiple of 8).
 DEV_INLINE bf16x8 lds_frag(const __bf16* tile, int row, int k0, int ldst) {
-  return *reinterpret_cast<const bf16x8*>(tile + row * ldst + k0);
+  return *reinterpret_cast<const bf16x8*>(tile + lds_rm_idx(row, k0, ldst));
 }
 
 // load an A fragment directly from global [S, hd]: lane row `grow`,
@@ -105,7 +133,7 @@ DEV_INLINE void tile_write_rows(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
     const int c = (slot % C8) * 8;
-    *reinterpret_cast<bf16x8*>(dst + row * ld + c) = r[i];
+    *reinterpret_cast<bf16x8*>(dst + lds_rm_idx(row, c, ld)) = r[i];
   }
 }
 
@@ -319,7 +347,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
       for (int n = 0; n < 4; ++n)
 #pragma unroll
         for (int reg = 0; reg < 4; ++reg)
-          pw[(kgrp * 4 + reg) * LDP + n * 16 + col] = (__bf16)p_val[n][reg];
+          pw[lds_rm_idx(kgrp * 4 + reg, n * 16 + col, LDP)] = (__bf16)p_val[n][reg];
 
       // PV: two K=32 steps over the 64-row kv tile
       __builtin_amdgcn_s_setprio(1);
@@ -494,7 +522,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg)
-        pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)ds_val[n][reg];
+        pw[lds_rm_idx(kgrp * 4 + reg, n * 16 + col, LDT)] = (__bf16)ds_val[n][reg];
 
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -655,8 +683,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     for (int n = 0; n < 4; ++n)
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        pw[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)pt_val[n][reg];
-        pw2[(kgrp * 4 + reg) * LDT + n * 16 + col] = (__bf16)dst_val[n][reg];
+        pw[lds_rm_idx(kgrp * 4 + reg, n * 16 + col, LDT)] = (__bf16)pt_val[n][reg];
+        pw2[lds_rm_idx(kgrp * 4 + reg, n * 16 + col, LDT)] = (__bf16)dst_val[n][reg];
       }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll

@@ -145,3 +145,44 @@ def test_rotation_preserves_layout_bijectivity():
     from lds_bank_model import tr64
     seen = {tr64(r, c) for r in range(128) for c in range(64)}
     assert len(seen) == 128 * 64
+
+
+def test_rowmajor_default_padding_writes_cf_reads_2way():
+    """The shipped default (8-element padding, no swizzle): staging writes
+    conflict-free, fragment reads 2-way."""
+    from lds_bank_model import (attn_k_tile_reads, attn_k_tile_writes,
+                                attn_p_tile_reads, attn_p_tile_writes)
+    for hd in (64, 128):
+        ok, w = verify(attn_k_tile_writes(hd, 8, False), "write_b128")
+        assert ok and w == 1, (hd, w)
+        _, r = verify(attn_k_tile_reads(hd, 8, False), "read_b128")
+        assert r == 2, (hd, r)
+    ok, w = verify(attn_p_tile_writes(8, False), "write_b16")
+    assert ok and w == 1, w
+    _, r = verify(attn_p_tile_reads(8, False), "read_b128")
+    assert r == 2, r
+
+
+def test_rowmajor_v2_swizzle_fully_conflict_free():
+    """RELORA_AMD_ROT_V2 row-major path (zero pad + per-row XOR): every K/P
+    write and read conflict-free at hd64 AND hd128 — and it saves the
+    padding LDS."""
+    from lds_bank_model import (attn_k_tile_reads, attn_k_tile_writes,
+                                attn_p_tile_reads, attn_p_tile_writes)
+    for hd in (64, 128):
+        ok, w = verify(attn_k_tile_writes(hd, 0, True), "write_b128")
+        assert ok and w == 1, (hd, "write", w)
+        ok, r = verify(attn_k_tile_reads(hd, 0, True), "read_b128")
+        assert ok and r == 1, (hd, "read", r)
+    ok, w = verify(attn_p_tile_writes(0, True), "write_b16")
+    assert ok and w == 1, w
+    ok, r = verify(attn_p_tile_reads(0, True), "read_b128")
+    assert ok and r == 1, r
+
+
+def test_rowmajor_swizzle_bijective():
+    from lds_bank_model import rm_swz
+    for ld in (64, 128):
+        seen = {(row * ld + col) ^ rm_swz(row, ld, True)
+                for row in range(64) for col in range(ld)}
+        assert len(seen) == 64 * ld

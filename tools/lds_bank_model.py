@@ -244,3 +244,61 @@ if __name__ == "__main__":
         row0s=range(0, 128, 16), k0s=(0, 32)), "read_b128")
     print(f"lora tr64 writes: {'conflict-free' if pt_ok else f'{pt_w}-way'};  "
           f"reads: {'conflict-free' if rd_ok else f'{rd_w}-way'}")
+
+
+# ---------------------------------------------------------------------------
+# row-major K/P tile patterns (attention.hip lds_frag / tile_write_rows /
+# the scalar P-writes), with optional per-row XOR swizzle — the
+# RELORA_AMD_ROT_V2 row-major path (zero padding + swizzle).
+# ---------------------------------------------------------------------------
+
+def rm_swz(row, ldst, enabled):
+    if not enabled:
+        return 0
+    return ((row & 15) if (ldst & 127) == 0 else (row & 7)) << 3
+
+
+def attn_k_tile_writes(hd, lpad, swz):
+    """tile_write_rows: bf16x8 (b128) writes at row*(hd+lpad) + c."""
+    ld = hd + lpad
+    c8 = hd // 8
+    nslot = TILE * c8
+    for i in range((nslot + 511) // 512):
+        for w in range(8):
+            addrs = {}
+            for lane in range(64):
+                slot = w * 64 + lane + i * 512
+                if slot >= nslot:
+                    continue
+                row, c = slot // c8, (slot % c8) * 8
+                addrs[lane] = 2 * ((row * ld + c) ^ rm_swz(row, ld, swz))
+            if addrs:
+                yield addrs
+
+
+def attn_k_tile_reads(hd, lpad, swz):
+    """lds_frag A-fragments: b128 reads at (n*16+col)*(hd+lpad) + k."""
+    ld = hd + lpad
+    for n in range(TILE // 16):
+        for kf in range(hd // 32):
+            yield {l: 2 * (((n * 16 + (l & 15)) * ld + kf * 32 + (l >> 4) * 8)
+                           ^ rm_swz(n * 16 + (l & 15), ld, swz))
+                   for l in range(64)}
+
+
+def attn_p_tile_writes(lpad, swz):
+    """scalar b16 P-writes at (kgrp*4+reg)*(64+lpad) + n*16+col."""
+    ld = TILE + lpad
+    for n in range(TILE // 16):
+        for reg in range(4):
+            yield {l: 2 * ((((l >> 4) * 4 + reg) * ld + n * 16 + (l & 15))
+                           ^ rm_swz((l >> 4) * 4 + reg, ld, swz))
+                   for l in range(64)}
+
+
+def attn_p_tile_reads(lpad, swz):
+    ld = TILE + lpad
+    for ks in range(TILE // 32):
+        yield {l: 2 * (((l & 15) * ld + ks * 32 + (l >> 4) * 8)
+                       ^ rm_swz(l & 15, ld, swz))
+               for l in range(64)}
