@@ -26,7 +26,30 @@
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// Default: 8-element row padding (staging writes conflict-free, fragment
+// reads 2-way per the bank model).  RELORA_AMD_ROT_V2: zero padding + a
+// per-row XOR swizzle on row-major tiles — conflict-free on every modeled
+// pattern for r % 64 == 0 (the flagship r=128 included); see
+// tools/lds_bank_model.py and tests/test_lds_bank_model.py.
+#ifdef RELORA_AMD_ROT_V2
+#define LPAD 0
+#else
 #define LPAD 8  // bf16 elements of LDS row padding (one 16B slot)
+#endif
+
+DEV_INLINE int rm_swz(int row, int ldst) {
+#ifdef RELORA_AMD_ROT_V2
+  return ((ldst & 127) == 0 ? (row & 15) : (row & 7)) << 3;
+#else
+  (void)row; (void)ldst;
+  return 0;
+#endif
+}
+// swizzled element index into a row-major [R][ldst] LDS tile — every
+// producer and consumer must address through this
+DEV_INLINE int rm_idx(int row, int col, int ldst) {
+  return (row * ldst + col) ^ rm_swz(row, ldst);
+}
 
 // Rotated layout for transposed LDS tiles with 64-element rows: element
 // (row, c) of an [R][64] image lives at row*64 + rot8(row, c), with the
@@ -147,7 +170,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
     } else {
       v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
-    *reinterpret_cast<bf16x8*>(p_im + row * ldt + c) = v;
+    *reinterpret_cast<bf16x8*>(p_im + rm_idx(row, c, ldt)) = v;
   }
   // stage Q tile as q_im[n][k]
   if (!TRANSQ) {
@@ -160,7 +183,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
       } else {
         v = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
       }
-      *reinterpret_cast<bf16x8*>(q_im + n * ldt + c) = v;
+      *reinterpret_cast<bf16x8*>(q_im + rm_idx(n, c, ldt)) = v;
     }
   } else {
     // Q is [r][N]: q_im[n][k] = Q[k][n0+n].  bf16x8 loads along n (contiguous
@@ -200,14 +223,14 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(p_im + (wr + mi * 16 + fr) * ldt + kk + kg);
+          *reinterpret_cast<const bf16x8*>(p_im + rm_idx(wr + mi * 16 + fr, kk + kg, ldt));
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int qrow = wc + ni * 16 + fr;
         const int koff = kk + kg;
         const bf16x8 b = *reinterpret_cast<const bf16x8*>(
             TRANSQ ? q_im + qrow * ldt + (koff & ~63) + rot8(qrow, koff & 63)
-                   : q_im + qrow * ldt + koff);
+                   : q_im + rm_idx(qrow, koff, ldt));
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mi][ni], 0, 0, 0);
       }
     }
@@ -227,7 +250,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
     for (int ni = 0; ni < 4; ++ni)
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        o_im[(wr + mi * 16 + crow + j) * OLD + wc + ni * 16 + fr] =
+        o_im[rm_idx(wr + mi * 16 + crow + j, wc + ni * 16 + fr, OLD)] =
             (__bf16)acc[mi][ni][j];
   __syncthreads();
 
@@ -237,7 +260,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
     const long m = m0 + row;
     const int n = n0 + c8;
     if (m >= M || n >= N) continue;
-    const __bf16* src_v = o_im + row * OLD + c8;
+    const __bf16* src_v = o_im + rm_idx(row, c8, OLD);
     const long flat = m * (long)N + n;  // mask bits are FLAT-packed over [M*N]
     __hip_bfloat16* o = out + flat;
     if (n + 8 <= N && (flat & 7) == 0) {

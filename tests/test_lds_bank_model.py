@@ -186,3 +186,60 @@ def test_rowmajor_swizzle_bijective():
         seen = {(row * ld + col) ^ rm_swz(row, ld, True)
                 for row in range(64) for col in range(ld)}
         assert len(seen) == 64 * ld
+
+
+def test_lora_rowmajor_v2_swizzle_conflict_free_at_r_multiple_64():
+    """lora_add (lora_skinny_kernel) row-major tiles under ROT_V2 (pad0 +
+    per-row XOR): staging writes, A-fragment reads, epilogue C-dump and
+    re-read all conflict-free at the flagship r=128 (and every r%64==0)."""
+    from lds_bank_model import access_cycles
+
+    def swz(row, ld):
+        return ((row & 15) if (ld & 127) == 0 else (row & 7)) << 3
+
+    for r in (64, 128, 192, 256):
+        ldt, r8, OLD = r, r // 8, 128
+
+        def p_stage():
+            total = 128 * r8
+            for it in range((total + 255) // 256):
+                for w in range(4):
+                    addrs = {}
+                    for lane in range(64):
+                        t = w * 64 + lane + it * 256
+                        if t >= total:
+                            continue
+                        row, c = t // r8, (t % r8) * 8
+                        addrs[lane] = 2 * ((row * ldt + c) ^ swz(row, ldt))
+                    if addrs:
+                        yield addrs
+
+        def a_reads():
+            for w in range(4):
+                wr = (w >> 1) * 64
+                for mi in range(4):
+                    for kk in range(0, r, 32):
+                        yield {l: 2 * (((wr + mi * 16 + (l & 15)) * ldt
+                                        + kk + (l >> 4) * 8)
+                                       ^ swz(wr + mi * 16 + (l & 15), ldt))
+                               for l in range(64)}
+
+        def o_reads():
+            total = 128 * 16
+            for it in range((total + 255) // 256):
+                for w in range(4):
+                    addrs = {}
+                    for lane in range(64):
+                        t = w * 64 + lane + it * 256
+                        if t >= total:
+                            continue
+                        row, c8 = t // 16, (t % 16) * 8
+                        addrs[lane] = 2 * ((row * OLD + c8) ^ swz(row, OLD))
+                    if addrs:
+                        yield addrs
+
+        for gen, kind in ((p_stage(), "write_b128"), (a_reads(), "read_b128"),
+                          (o_reads(), "read_b128")):
+            for addrs in gen:
+                c, m = access_cycles(addrs, kind)
+                assert c == m, (r, kind, c, m)
