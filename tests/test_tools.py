@@ -129,3 +129,15 @@ def test_tools_run_as_scripts(tiny_run, tmp_path):
         capture_output=True, text=True, env=env, timeout=300)
     assert r.returncode == 0, r.stderr
     assert (tmp_path / "spectra.png").exists()
+
+
+def test_search_lds_layout_tool():
+    """The layout searcher verifies Q_V2 as a zero-conflict solution and
+    scores the current rotation's residual conflicts."""
+    import subprocess
+    env = {**os.environ, "PYTHONPATH": REPO}
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "search_lds_layout.py")],
+        capture_output=True, text=True, env=env, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "Q_V2          : 0" in r.stdout
