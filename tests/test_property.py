@@ -114,3 +114,39 @@ def test_cosine_restarts_schedule_invariants(restart_every, cycles, warmup,
         seg = lrs[b:min(b + restart_warmup, total)]
         if len(seg) >= 2:
             assert np.all(np.diff(seg) >= -1e-6), (c, seg)
+
+
+@settings(max_examples=25, deadline=None, derandomize=True)
+@given(
+    docs=st.lists(st.lists(st.integers(0, 60000), min_size=1, max_size=50),
+                  min_size=1, max_size=20),
+    vocab=st.sampled_from([100, 40000, 70000]),
+)
+def test_mmap_indexed_dataset_roundtrip_and_slices(tmp_path_factory, docs, vocab):
+    """MMapIndexedDataset: arbitrary corpora round-trip bit-exactly and
+    get(idx, offset, length) windows match the source (the contract
+    GPT2Dataset sample addressing relies on)."""
+    import numpy as np
+
+    from relora_amd.data import indexed_dataset as idx
+
+    tmp = tmp_path_factory.mktemp("mmapprop")
+    prefix = str(tmp / "corpus")
+    dtype = idx.best_fitting_dtype(vocab)
+    docs = [[t % vocab for t in d] for d in docs]
+    builder = idx.make_builder(prefix + ".bin", "mmap", vocab_size=vocab)
+    for d in docs:
+        builder.add_item(torch.tensor(d, dtype=torch.int64))
+        builder.end_document()
+    builder.finalize(prefix + ".idx")
+
+    ds = idx.make_dataset(prefix, "mmap", skip_warmup=True)
+    assert len(ds) == len(docs)
+    assert ds[0].dtype == np.dtype(dtype)
+    for i, d in enumerate(docs):
+        np.testing.assert_array_equal(ds[i], np.asarray(d))
+        # offset/length windows
+        if len(d) > 1:
+            win = ds.get(i, offset=1, length=len(d) - 1)
+            np.testing.assert_array_equal(win, np.asarray(d[1:]))
+    np.testing.assert_array_equal(ds.sizes, [len(d) for d in docs])

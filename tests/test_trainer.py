@@ -367,3 +367,20 @@ def test_regime_quantized_relora(tmp_path):
     state = json.load(open(ckpt / "training_state.json"))
     assert state["update_step"] == 6
     assert state["n_lora_restarts"] >= 1
+
+
+def test_args_max_train_tokens():
+    """--max_train_tokens accepts 1M/1B suffixes and derives
+    num_training_steps = tokens // total_batch_size (reference
+    args_utils.py:49-51, training_utils.max_train_tokens_to_number)."""
+    args = parse_args([
+        "--synthetic_data", "true", "--batch_size", "2",
+        "--total_batch_size", "4", "--max_train_tokens", "2M",
+    ])
+    assert args.max_train_tokens == 2_000_000
+    assert args.num_training_steps == 2_000_000 // 4
+    args = parse_args([
+        "--synthetic_data", "true", "--batch_size", "2",
+        "--total_batch_size", "4", "--max_train_tokens", "1B",
+    ])
+    assert args.max_train_tokens == 1_000_000_000
