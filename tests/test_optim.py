@@ -116,3 +116,31 @@ def test_get_last_training_state(tmp_path):
     state, ckpt = get_last_training_state(str(tmp_path))
     assert ckpt.endswith("model_25")
     assert state["update_step"] == 25
+
+
+def test_print_optimizer_state_size(capsys):
+    """Counts Adam moment floats; understands the ZeRO `.optim` indirection
+    (reference training_utils.py:367-388)."""
+    from relora_amd.training_utils import print_optimizer_state_size
+
+    p = torch.nn.Parameter(torch.randn(100, 10))
+    opt = torch.optim.Adam([p], lr=1e-3)
+    p.grad = torch.randn_like(p)
+    opt.step()
+    print_optimizer_state_size(opt)
+    out = capsys.readouterr().out
+    assert "first moment" in out and "0.00M" in out  # 1000 floats = 0.00M
+
+
+def test_check_lr_and_alert(capfd):
+    """Warns (and wandb-alerts) when post-reset lr exceeds the bound
+    (reference training_utils.py:391-404).  The rank-0 logger writes to the
+    stderr fd, so capture at fd level."""
+    from relora_amd.training_utils import check_lr_and_alert
+
+    p = torch.nn.Parameter(torch.zeros(1))
+    opt = torch.optim.SGD([p], lr=0.1)
+    check_lr_and_alert(opt, max_lr=1.0)   # fine: no warning
+    assert "lr after the reset" not in capfd.readouterr().err
+    check_lr_and_alert(opt, max_lr=0.01)  # too large: warns
+    assert "lr after the reset" in capfd.readouterr().err
