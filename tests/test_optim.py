@@ -132,15 +132,23 @@ def test_print_optimizer_state_size(capsys):
     assert "first moment" in out and "0.00M" in out  # 1000 floats = 0.00M
 
 
-def test_check_lr_and_alert(capfd):
+def test_check_lr_and_alert():
     """Warns (and wandb-alerts) when post-reset lr exceeds the bound
-    (reference training_utils.py:391-404).  The rank-0 logger writes to the
-    stderr fd, so capture at fd level."""
-    from relora_amd.training_utils import check_lr_and_alert
+    (reference training_utils.py:391-404).  Captures by swapping the
+    logger's handler stream (it binds stderr at import time)."""
+    import io
 
-    p = torch.nn.Parameter(torch.zeros(1))
-    opt = torch.optim.SGD([p], lr=0.1)
-    check_lr_and_alert(opt, max_lr=1.0)   # fine: no warning
-    assert "lr after the reset" not in capfd.readouterr().err
-    check_lr_and_alert(opt, max_lr=0.01)  # too large: warns
-    assert "lr after the reset" in capfd.readouterr().err
+    from relora_amd.training_utils import check_lr_and_alert
+    from relora_amd.utils.logging import logger
+
+    buf = io.StringIO()
+    old = logger._handler.setStream(buf)
+    try:
+        p = torch.nn.Parameter(torch.zeros(1))
+        opt = torch.optim.SGD([p], lr=0.1)
+        check_lr_and_alert(opt, max_lr=1.0)   # fine: no warning
+        assert "lr after the reset" not in buf.getvalue()
+        check_lr_and_alert(opt, max_lr=0.01)  # too large: warns
+        assert "lr after the reset" in buf.getvalue()
+    finally:
+        logger._handler.setStream(old)
