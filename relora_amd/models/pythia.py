@@ -263,7 +263,8 @@ class GPTNeoXModel(GPTNeoXPreTrainedModel):
 
 
 class GPTNeoXForCausalLM(GPTNeoXPreTrainedModel):
-    _tied_weights_keys = ["embed_out.weight"]
+    # dict form (target -> source) per the installed transformers' tying API
+    _tied_weights_keys = {"embed_out.weight": "gpt_neox.embed_in.weight"}
 
     def __init__(self, config):
         super().__init__(config)

@@ -211,3 +211,33 @@ def test_pythia_rope_scaling_variants():
             outs[label] = m(input_ids=x).logits
     assert not torch.allclose(outs["base"], outs["linear"])
     assert not torch.allclose(outs["base"], outs["dynamic"])
+
+
+def test_pythia_tied_embeddings():
+    """tie_word_embeddings=True shares embed_in/embed_out storage (the
+    pythia tied-head variant; reference modeling_pythia.py head wiring),
+    gradients flow, and the untied default does not share."""
+    import torch as _t
+
+    from relora_amd.models.pythia import GPTNeoXConfig, GPTNeoXForCausalLM
+
+    base = dict(vocab_size=128, hidden_size=32, num_hidden_layers=2,
+                num_attention_heads=4, intermediate_size=64,
+                max_position_embeddings=64, rotary_pct=0.25)
+    tied = GPTNeoXForCausalLM(GPTNeoXConfig(**base, tie_word_embeddings=True))
+    assert (tied.embed_out.weight.data_ptr()
+            == tied.gpt_neox.embed_in.weight.data_ptr())
+    untied = GPTNeoXForCausalLM(GPTNeoXConfig(**base, tie_word_embeddings=False))
+    assert (untied.embed_out.weight.data_ptr()
+            != untied.gpt_neox.embed_in.weight.data_ptr())
+
+    x = _t.randint(0, 128, (2, 16))
+    loss = tied(input_ids=x, labels=x).loss
+    loss.backward()
+    assert tied.gpt_neox.embed_in.weight.grad is not None
+    # save/load keeps the tie
+    sd = tied.state_dict()
+    reloaded = GPTNeoXForCausalLM(GPTNeoXConfig(**base, tie_word_embeddings=True))
+    reloaded.load_state_dict(sd)
+    assert (reloaded.embed_out.weight.data_ptr()
+            == reloaded.gpt_neox.embed_in.weight.data_ptr())
