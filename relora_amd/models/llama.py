@@ -240,6 +240,10 @@ class LlamaModel(LlamaPreTrainedModel):
 
 
 class LlamaForCausalLM(LlamaPreTrainedModel):
+    # honored only when config.tie_word_embeddings is True (the reference's
+    # llama recipes keep the head untied — configs/llama_*.json)
+    _tied_weights_keys = {"lm_head.weight": "model.embed_tokens.weight"}
+
     def __init__(self, config):
         super().__init__(config)
         self.model = LlamaModel(config)

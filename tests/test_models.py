@@ -241,3 +241,23 @@ def test_pythia_tied_embeddings():
     reloaded.load_state_dict(sd)
     assert (reloaded.embed_out.weight.data_ptr()
             == reloaded.gpt_neox.embed_in.weight.data_ptr())
+
+
+def test_llama_tie_word_embeddings_honored():
+    """tie_word_embeddings=True shares lm_head/embed_tokens storage; the
+    default (and every shipped configs/llama_*.json) stays untied."""
+    import torch as _t
+
+    from relora_amd.models.config import LlamaConfig
+    from relora_amd.models.llama import LlamaForCausalLM
+
+    base = dict(vocab_size=128, hidden_size=32, intermediate_size=64,
+                num_hidden_layers=2, num_attention_heads=4,
+                max_position_embeddings=64)
+    for tie in (True, False):
+        m = LlamaForCausalLM(LlamaConfig(**base, tie_word_embeddings=tie))
+        shared = (m.lm_head.weight.data_ptr()
+                  == m.model.embed_tokens.weight.data_ptr())
+        assert shared == tie
+        x = _t.randint(0, 128, (1, 8))
+        m(input_ids=x, labels=x).loss.backward()
