@@ -20,6 +20,7 @@ from transformers.modeling_outputs import (
     CausalLMOutputWithPast,
     SequenceClassifierOutputWithPast,
 )
+from transformers.generation import GenerationMixin
 from transformers.modeling_utils import PreTrainedModel
 
 from relora_amd import ops
@@ -207,6 +208,11 @@ class LlamaModel(LlamaPreTrainedModel):
         hidden_states = inputs_embeds
         use_cache = bool(use_cache) and not self.gradient_checkpointing
 
+        # `generate` passes a Cache object; internally we speak legacy tuples
+        from relora_amd.models.cache_compat import cache_like, cache_to_legacy
+        cache_template = past_key_values
+        past_key_values = cache_to_legacy(past_key_values)
+
         all_hidden_states = [] if output_hidden_states else None
         next_cache = [] if use_cache else None
         for i, layer in enumerate(self.layers):
@@ -230,16 +236,17 @@ class LlamaModel(LlamaPreTrainedModel):
         if output_hidden_states:
             all_hidden_states.append(hidden_states)
 
+        out_cache = cache_like(next_cache, cache_template)
         if not return_dict:
-            return tuple(v for v in (hidden_states, next_cache, all_hidden_states) if v is not None)
+            return tuple(v for v in (hidden_states, out_cache, all_hidden_states) if v is not None)
         return BaseModelOutputWithPast(
             last_hidden_state=hidden_states,
-            past_key_values=tuple(next_cache) if next_cache else None,
+            past_key_values=out_cache,
             hidden_states=tuple(all_hidden_states) if all_hidden_states else None,
         )
 
 
-class LlamaForCausalLM(LlamaPreTrainedModel):
+class LlamaForCausalLM(LlamaPreTrainedModel, GenerationMixin):
     # honored only when config.tie_word_embeddings is True (the reference's
     # llama recipes keep the head untied — configs/llama_*.json)
     _tied_weights_keys = {"lm_head.weight": "model.embed_tokens.weight"}
@@ -321,8 +328,17 @@ class LlamaForCausalLM(LlamaPreTrainedModel):
         )
 
     def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
-        if past_key_values:
-            input_ids = input_ids[:, -1:]
+        # trim by the ACTUAL cached length: `generate` pre-creates a
+        # DynamicCache whose layers are initialized but empty, so truthiness
+        # of the cache object is not "has tokens"
+        past_len = 0
+        if past_key_values is not None:
+            if hasattr(past_key_values, "get_seq_length"):
+                past_len = int(past_key_values.get_seq_length())
+            elif past_key_values:
+                past_len = past_key_values[0][0].shape[-2]
+        if past_len:
+            input_ids = input_ids[:, past_len:]
         return {
             "input_ids": input_ids,
             "past_key_values": past_key_values,

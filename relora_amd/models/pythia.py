@@ -16,6 +16,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 import torch.utils.checkpoint
 from transformers.modeling_outputs import BaseModelOutputWithPast, CausalLMOutputWithPast
+from transformers.generation import GenerationMixin
 from transformers.modeling_utils import PreTrainedModel
 
 from relora_amd import ops
@@ -227,6 +228,11 @@ class GPTNeoXModel(GPTNeoXPreTrainedModel):
         hidden_states = self.emb_dropout(inputs_embeds)
         use_cache = bool(use_cache) and not self.gradient_checkpointing
 
+        # `generate` passes a Cache object; internally we speak legacy tuples
+        from relora_amd.models.cache_compat import cache_like, cache_to_legacy
+        cache_template = past_key_values
+        past_key_values = cache_to_legacy(past_key_values)
+
         all_hidden_states = [] if output_hidden_states else None
         next_cache = [] if use_cache else None
         for i, layer in enumerate(self.layers):
@@ -253,16 +259,17 @@ class GPTNeoXModel(GPTNeoXPreTrainedModel):
         if output_hidden_states:
             all_hidden_states.append(hidden_states)
 
+        out_cache = cache_like(next_cache, cache_template)
         if not return_dict:
-            return tuple(v for v in (hidden_states, next_cache, all_hidden_states) if v is not None)
+            return tuple(v for v in (hidden_states, out_cache, all_hidden_states) if v is not None)
         return BaseModelOutputWithPast(
             last_hidden_state=hidden_states,
-            past_key_values=tuple(next_cache) if next_cache else None,
+            past_key_values=out_cache,
             hidden_states=tuple(all_hidden_states) if all_hidden_states else None,
         )
 
 
-class GPTNeoXForCausalLM(GPTNeoXPreTrainedModel):
+class GPTNeoXForCausalLM(GPTNeoXPreTrainedModel, GenerationMixin):
     # dict form (target -> source) per the installed transformers' tying API
     _tied_weights_keys = {"embed_out.weight": "gpt_neox.embed_in.weight"}
 
@@ -338,8 +345,17 @@ class GPTNeoXForCausalLM(GPTNeoXPreTrainedModel):
         )
 
     def prepare_inputs_for_generation(self, input_ids, past_key_values=None, **kwargs):
-        if past_key_values:
-            input_ids = input_ids[:, -1:]
+        # trim by the ACTUAL cached length: `generate` pre-creates a
+        # DynamicCache whose layers are initialized but empty, so truthiness
+        # of the cache object is not "has tokens"
+        past_len = 0
+        if past_key_values is not None:
+            if hasattr(past_key_values, "get_seq_length"):
+                past_len = int(past_key_values.get_seq_length())
+            elif past_key_values:
+                past_len = past_key_values[0][0].shape[-2]
+        if past_len:
+            input_ids = input_ids[:, past_len:]
         return {
             "input_ids": input_ids,
             "past_key_values": past_key_values,

@@ -261,3 +261,37 @@ def test_llama_tie_word_embeddings_honored():
         assert shared == tie
         x = _t.randint(0, 128, (1, 8))
         m(input_ids=x, labels=x).loss.backward()
+
+
+def test_generate_matches_manual_greedy():
+    """model.generate (GenerationMixin + DynamicCache under the installed
+    transformers) produces exactly the tokens of a manual greedy
+    full-recompute loop, for both model families."""
+    import torch as _t
+
+    from relora_amd.models.config import GPTNeoXConfig, LlamaConfig
+    from relora_amd.models.llama import LlamaForCausalLM
+    from relora_amd.models.pythia import GPTNeoXForCausalLM
+
+    _t.manual_seed(0)
+    x = _t.randint(2, 128, (1, 5))
+    models = [
+        LlamaForCausalLM(LlamaConfig(
+            vocab_size=128, hidden_size=32, intermediate_size=64,
+            num_hidden_layers=2, num_attention_heads=4,
+            max_position_embeddings=64, bos_token_id=0, eos_token_id=1)),
+        GPTNeoXForCausalLM(GPTNeoXConfig(
+            vocab_size=128, hidden_size=32, num_hidden_layers=2,
+            num_attention_heads=4, intermediate_size=64,
+            max_position_embeddings=64, rotary_pct=0.25,
+            bos_token_id=0, eos_token_id=1)),
+    ]
+    for m in models:
+        m.eval()
+        out = m.generate(x, max_new_tokens=6, do_sample=False)
+        cur = x.clone()
+        for _ in range(6):
+            with _t.no_grad():
+                cur = _t.cat([cur, m(input_ids=cur).logits[:, -1:].argmax(-1)],
+                             dim=1)
+        assert _t.equal(out, cur), type(m).__name__
