@@ -105,8 +105,12 @@ class ReLoRaModel(torch.nn.Module):
             use_double_quant=use_double_quant,
         )
 
-        # expose the wrapped model's forward (reference relora.py:89)
+        # expose the wrapped model's forward (reference relora.py:89) and,
+        # beyond the reference, its generation surface so a wrapped model
+        # can decode directly
         self.forward = self.wrapped_model.forward
+        if hasattr(self.wrapped_model, "generate"):
+            self.generate = self.wrapped_model.generate
 
         target_modules_list = [target_modules] if isinstance(target_modules, str) else target_modules
 

@@ -250,3 +250,30 @@ def test_quantized_save_load_roundtrip(tiny_llama_config, tmp_path):
         a = wrapped(input_ids=x, labels=x).loss
         b = reloaded(input_ids=x, labels=x).loss
     assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_relora_wrapped_generate(tiny_llama_config):
+    """A ReLoRA-wrapped model can decode directly (generate delegates to
+    the wrapped model; adapters participate in the forward)."""
+    import torch as _t
+
+    from relora_amd.models.llama import LlamaForCausalLM
+    from relora_amd.relora import ReLoRaModel
+
+    _t.manual_seed(0)
+    m = LlamaForCausalLM(tiny_llama_config)
+    w = ReLoRaModel(m, r=4, lora_alpha=8, lora_dropout=0.0,
+                    target_modules=["attn", "attention", "mlp"],
+                    keep_original_weights=True)
+    w.eval()
+    x = _t.randint(2, tiny_llama_config.vocab_size, (1, 4))
+    out = w.generate(x, max_new_tokens=4, do_sample=False)
+    assert out.shape == (1, 8)
+    # adapters do contribute: perturb both factors (lora_A is zero-init
+    # under keep_original_weights, so B alone is inert), output changes
+    with _t.no_grad():
+        for n, p in w.named_parameters():
+            if "lora_A" in n or "lora_B" in n:
+                p.add_(0.5)
+    out2 = w.generate(x, max_new_tokens=4, do_sample=False)
+    assert not _t.equal(out, out2)
