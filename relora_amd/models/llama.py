@@ -49,15 +49,28 @@ class LlamaRotaryEmbedding(nn.Module):
         self.register_buffer("cos_cached", cos, persistent=False)
         self.register_buffer("sin_cached", sin, persistent=False)
 
+    def _cache_is_valid(self):
+        # from_pretrained materializes the module from the meta device, which
+        # leaves non-persistent buffers UNINITIALIZED; cos(position 0) == 1
+        # for every frequency in a real cache, so verify once per process
+        if getattr(self, "_cache_checked", False):
+            return True
+        ok = (self.cos_cached.dtype == torch.float32
+              and bool((self.cos_cached[0] == 1).all()))
+        self._cache_checked = ok
+        return ok
+
     def forward(self, x, seq_len):
-        # Rebuild if the cache grew or a model-wide .to(dtype) cast it away
-        # from fp32 (the RoPE kernel consumes fp32 tables).
-        if seq_len > self.max_seq_len_cached or self.cos_cached.dtype != torch.float32:
+        # Rebuild if the cache grew, a model-wide .to(dtype) cast it away
+        # from fp32 (the RoPE kernel consumes fp32 tables), or meta-device
+        # materialization left it uninitialized.
+        if seq_len > self.max_seq_len_cached or not self._cache_is_valid():
             self.max_seq_len_cached = max(seq_len, self.max_seq_len_cached)
             cos, sin = ops.build_rope_cache(
                 self.dim, self.max_seq_len_cached, self.base, x.device)
             self.register_buffer("cos_cached", cos, persistent=False)
             self.register_buffer("sin_cached", sin, persistent=False)
+            self._cache_checked = True
         return self.cos_cached.to(x.device), self.sin_cached.to(x.device)
 
 

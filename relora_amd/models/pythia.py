@@ -64,13 +64,26 @@ class GPTNeoXRotary(nn.Module):
         self.register_buffer("sin_cached", emb.sin(), persistent=False)
         self.max_seq_len_cached = seq_len
 
+    def _cache_is_valid(self):
+        # from_pretrained materializes from the meta device, leaving
+        # non-persistent buffers uninitialized; cos(position 0) == 1 in any
+        # real cache — verify once per process
+        if getattr(self, "_cache_checked", False):
+            return True
+        ok = (self.cos_cached.dtype == torch.float32
+              and bool((self.cos_cached[0] == 1).all()))
+        self._cache_checked = ok
+        return ok
+
     def forward(self, x, seq_len):
         if seq_len > self.max_seq_len_cached or (
             self.scaling_type == "dynamic" and seq_len != self.max_seq_len_cached
-        ) or self.cos_cached.dtype != torch.float32:
+        ) or not self._cache_is_valid():
             # fp32 tables are part of the RoPE kernel contract; a model-wide
-            # .to(dtype) may have cast the buffers.
+            # .to(dtype) may have cast the buffers, and meta-device loading
+            # may have voided them.
             self._build(max(seq_len, self.max_position_embeddings), x.device)
+            self._cache_checked = True
         return self.cos_cached.to(x.device), self.sin_cached.to(x.device)
 
 

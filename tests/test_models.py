@@ -295,3 +295,40 @@ def test_generate_matches_manual_greedy():
                 cur = _t.cat([cur, m(input_ids=cur).logits[:, -1:].argmax(-1)],
                              dim=1)
         assert _t.equal(out, cur), type(m).__name__
+
+
+def test_hf_from_pretrained_classmethod_exact():
+    """PreTrainedModel.from_pretrained on our classes is bit-exact.
+    Regression: transformers materializes models from the META device, which
+    voids non-persistent buffers — the RoPE caches must detect that and
+    rebuild (they silently held garbage before; logits were off by ~1e-2)."""
+    import tempfile
+
+    import torch as _t
+
+    from relora_amd.models.config import GPTNeoXConfig, LlamaConfig
+    from relora_amd.models.llama import LlamaForCausalLM
+    from relora_amd.models.pythia import GPTNeoXForCausalLM
+    from relora_amd.utils.checkpoint import save_pretrained_compat
+
+    _t.manual_seed(0)
+    x = _t.randint(0, 128, (1, 8))
+    cases = [
+        (LlamaForCausalLM, LlamaConfig(
+            vocab_size=128, hidden_size=32, intermediate_size=64,
+            num_hidden_layers=2, num_attention_heads=4,
+            max_position_embeddings=64)),
+        (GPTNeoXForCausalLM, GPTNeoXConfig(
+            vocab_size=128, hidden_size=32, num_hidden_layers=2,
+            num_attention_heads=4, intermediate_size=64,
+            max_position_embeddings=64, rotary_pct=0.25)),
+    ]
+    for cls, cfg in cases:
+        m = cls(cfg).eval()
+        d = tempfile.mkdtemp()
+        save_pretrained_compat(m, d)
+        m2 = cls.from_pretrained(d).eval()
+        with _t.no_grad():
+            a = m(input_ids=x).logits
+            b = m2(input_ids=x).logits
+        assert _t.equal(a, b), (cls.__name__, (a - b).abs().max())
