@@ -52,11 +52,12 @@ class LlamaRotaryEmbedding(nn.Module):
     def _cache_is_valid(self):
         # from_pretrained materializes the module from the meta device, which
         # leaves non-persistent buffers UNINITIALIZED; cos(position 0) == 1
-        # for every frequency in a real cache, so verify once per process
+        # for every frequency in a real cache, so verify once per process.
+        # (The fp32-dtype check stays SEPARATE in forward: a model-wide
+        # .to(dtype) can cast the cache at any later point.)
         if getattr(self, "_cache_checked", False):
             return True
-        ok = (self.cos_cached.dtype == torch.float32
-              and bool((self.cos_cached[0] == 1).all()))
+        ok = bool((self.cos_cached[0] == 1).all())
         self._cache_checked = ok
         return ok
 
@@ -64,7 +65,9 @@ class LlamaRotaryEmbedding(nn.Module):
         # Rebuild if the cache grew, a model-wide .to(dtype) cast it away
         # from fp32 (the RoPE kernel consumes fp32 tables), or meta-device
         # materialization left it uninitialized.
-        if seq_len > self.max_seq_len_cached or not self._cache_is_valid():
+        if (seq_len > self.max_seq_len_cached
+                or self.cos_cached.dtype != torch.float32
+                or not self._cache_is_valid()):
             self.max_seq_len_cached = max(seq_len, self.max_seq_len_cached)
             cos, sin = ops.build_rope_cache(
                 self.dim, self.max_seq_len_cached, self.base, x.device)

@@ -67,18 +67,18 @@ class GPTNeoXRotary(nn.Module):
     def _cache_is_valid(self):
         # from_pretrained materializes from the meta device, leaving
         # non-persistent buffers uninitialized; cos(position 0) == 1 in any
-        # real cache — verify once per process
+        # real cache — verify once per process.  (The fp32-dtype check stays
+        # SEPARATE in forward: a .to(dtype) can cast the cache later.)
         if getattr(self, "_cache_checked", False):
             return True
-        ok = (self.cos_cached.dtype == torch.float32
-              and bool((self.cos_cached[0] == 1).all()))
+        ok = bool((self.cos_cached[0] == 1).all())
         self._cache_checked = ok
         return ok
 
     def forward(self, x, seq_len):
         if seq_len > self.max_seq_len_cached or (
             self.scaling_type == "dynamic" and seq_len != self.max_seq_len_cached
-        ) or not self._cache_is_valid():
+        ) or self.cos_cached.dtype != torch.float32 or not self._cache_is_valid():
             # fp32 tables are part of the RoPE kernel contract; a model-wide
             # .to(dtype) may have cast the buffers, and meta-device loading
             # may have voided them.

@@ -332,3 +332,26 @@ def test_hf_from_pretrained_classmethod_exact():
             a = m(input_ids=x).logits
             b = m2(input_ids=x).logits
         assert _t.equal(a, b), (cls.__name__, (a - b).abs().max())
+
+
+def test_rope_cache_rebuilds_on_cast_after_forward():
+    """The forward-then-cast order: a model that already ran (validity flag
+    set) and is THEN cast to bf16 must still rebuild its fp32 RoPE tables
+    on the next forward (the kernel contract)."""
+    import torch as _t
+
+    from relora_amd.models.config import LlamaConfig
+    from relora_amd.models.llama import LlamaForCausalLM
+
+    m = LlamaForCausalLM(LlamaConfig(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=4,
+        max_position_embeddings=64)).eval()
+    x = _t.randint(0, 128, (1, 8))
+    with _t.no_grad():
+        m(input_ids=x)                      # sets the validity flag
+    m = m.to(_t.bfloat16)                   # casts the cached tables
+    with _t.no_grad():
+        m(input_ids=x)
+    rot = m.model.layers[0].self_attn.rotary_emb
+    assert rot.cos_cached.dtype == _t.float32
