@@ -55,7 +55,7 @@ class _HipRMSNorm(torch.autograd.Function):
 
 
 def rmsnorm(x, weight, eps=1e-6):
-    if hip.use_hip(x):
+    if hip.use_hip(x, "rmsnorm"):
         return _HipRMSNorm.apply(x, weight, eps)
     return rmsnorm_torch(x, weight, eps)
 
@@ -83,7 +83,7 @@ class _HipLayerNorm(torch.autograd.Function):
 
 
 def layernorm(x, weight, bias, eps=1e-5):
-    if hip.use_hip(x):
+    if hip.use_hip(x, "layernorm"):
         return _HipLayerNorm.apply(x, weight, bias, eps)
     return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
 
@@ -156,7 +156,7 @@ class _HipRoPE(torch.autograd.Function):
 def rope(q, k, cos, sin, position_ids=None):
     """Apply rotary embedding to q and k. position_ids only supported on the
     torch path (training uses the contiguous [0..S) default)."""
-    if position_ids is None and hip.use_hip(q):
+    if position_ids is None and hip.use_hip(q, "rope"):
         return _HipRoPE.apply(q, k, cos, sin)
     return rope_torch(q, k, cos, sin, position_ids)
 
@@ -187,7 +187,7 @@ class _HipSwiGLU(torch.autograd.Function):
 
 
 def swiglu(gate, up):
-    if hip.use_hip(gate):
+    if hip.use_hip(gate, "swiglu"):
         return _HipSwiGLU.apply(gate, up)
     return swiglu_torch(gate, up)
 
@@ -224,7 +224,7 @@ def flash_attention(q, k, v, causal=True, dropout_p=0.0, scale=None):
     """q,k,v: [B, nh, S, hd] -> [B, nh, S, hd]."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
-    if causal and dropout_p == 0.0 and hip.use_hip(q):
+    if causal and dropout_p == 0.0 and hip.use_hip(q, "attention"):
         return _HipFlashAttention.apply(q, k, v, scale)
     return F.scaled_dot_product_attention(
         q, k, v, dropout_p=dropout_p, is_causal=causal, scale=scale
@@ -267,7 +267,7 @@ class _FusedCrossEntropy(torch.autograd.Function):
         loss_sum = torch.zeros((), dtype=torch.float32, device=device)
         valid = labels != ignore_index
         n_valid = int(valid.sum().item())
-        use_hip = hip.use_hip(hidden)
+        use_hip = hip.use_hip(hidden, "ce")
         keep_logits = _CE_SAVE_LOGITS and M <= _CE_CHUNK
         saved_logits = None
         for s in range(0, M, _CE_CHUNK):
@@ -295,7 +295,7 @@ class _FusedCrossEntropy(torch.autograd.Function):
         hidden, weight, labels, lse_all, saved_logits = ctx.saved_tensors
         ignore_index = ctx.ignore_index
         M, H = hidden.shape
-        use_hip = hip.use_hip(hidden)
+        use_hip = hip.use_hip(hidden, "ce")
         dh = torch.empty_like(hidden)
         single_chunk = M <= _CE_CHUNK
         dw_acc = None if single_chunk else torch.zeros(
@@ -429,7 +429,7 @@ def _fused_ok(x, weight, lora_A, scale, lora_only):
     # odd in/out dims (llama_1b intermediate 5461) are CORRECT through the
     # fused kernels (alignment-guarded fallbacks) but measured slower than
     # the hipBLASLt composition — keep them on the library path
-    return (hip.use_hip(x) and not lora_only and not torch.is_tensor(scale)
+    return (hip.use_hip(x, "lora") and not lora_only and not torch.is_tensor(scale)
             and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
             and r % 32 == 0 and r <= 256
             and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0

@@ -46,13 +46,31 @@ def allow_fallback():
     return os.environ.get("RELORA_AMD_ALLOW_FALLBACK", "0") == "1"
 
 
-def use_hip(t):
+_disabled_ops = None
+
+
+def _op_disabled(op):
+    """RELORA_AMD_DISABLE_OPS='attention,rmsnorm,...' routes the named ops
+    through the torch composition on GPU — an explicit per-kernel A/B
+    switch for measurement, never a silent fallback (the env must be set
+    deliberately).  Parsed once per process."""
+    global _disabled_ops
+    if _disabled_ops is None:
+        raw = os.environ.get("RELORA_AMD_DISABLE_OPS", "")
+        _disabled_ops = {s.strip() for s in raw.split(",") if s.strip()}
+    return op is not None and op in _disabled_ops
+
+
+def use_hip(t, op=None):
     """True if op dispatch should take the HIP kernel path for tensor `t`.
 
     Raises when `t` is on GPU and the extension is missing (fail loudly —
     a silent eager fallback on the GPU box would invalidate benchmarks).
+    `op` names the kernel for the RELORA_AMD_DISABLE_OPS A/B switch.
     """
     if not t.is_cuda:
+        return False
+    if _op_disabled(op):
         return False
     _try_load()
     if _ext is not None:

@@ -55,7 +55,7 @@ class AdamW(torch.optim.Optimizer):
             beta1, beta2 = group["betas"]
             lr, eps, wd = group["lr"], group["eps"], group["weight_decay"]
 
-            if params[0].is_cuda and hip.use_hip(params[0]):
+            if params[0].is_cuda and hip.use_hip(params[0], "adamw"):
                 # group by identical step count (true except after partial loads)
                 by_step = {}
                 for i, s in enumerate(steps):
@@ -99,7 +99,7 @@ def clip_grad_norm_(parameters, max_norm, norm_type=2.0, error_if_nonfinite=True
     assert norm_type == 2.0, "only L2 clipping is supported"
     device = grads[0].device
 
-    if grads[0].is_cuda and hip.use_hip(grads[0]):
+    if grads[0].is_cuda and hip.use_hip(grads[0], "clip"):
         total_norm = hip.ext().multi_tensor_l2norm(grads)
     else:
         total_norm = torch.linalg.vector_norm(
@@ -113,7 +113,7 @@ def clip_grad_norm_(parameters, max_norm, norm_type=2.0, error_if_nonfinite=True
         )
     clip_coef = max_norm / (total_norm + 1e-6)
     if clip_coef < 1:
-        if grads[0].is_cuda and hip.use_hip(grads[0]):
+        if grads[0].is_cuda and hip.use_hip(grads[0], "clip"):
             hip.ext().multi_tensor_scale_(grads, float(clip_coef))
         else:
             torch._foreach_mul_(grads, clip_coef.to(device))
