@@ -355,3 +355,30 @@ def test_rope_cache_rebuilds_on_cast_after_forward():
         m(input_ids=x)
     rot = m.model.layers[0].self_attn.rotary_emb
     assert rot.cos_cached.dtype == _t.float32
+
+
+def test_cache_compat_unit():
+    """cache_compat: legacy passthrough, empty-cache handling, and cache
+    reconstruction in the template's type."""
+    import torch as _t
+    from transformers.cache_utils import DynamicCache
+
+    from relora_amd.models.cache_compat import cache_like, cache_to_legacy
+
+    assert cache_to_legacy(None) is None
+    assert cache_to_legacy(()) is None
+    legacy = ((_t.zeros(1, 2, 3, 4), _t.ones(1, 2, 3, 4)),)
+    assert cache_to_legacy(legacy) is legacy
+    assert cache_to_legacy(DynamicCache()) is None  # fresh/empty
+
+    filled = DynamicCache()
+    filled.update(_t.zeros(1, 2, 3, 4), _t.ones(1, 2, 3, 4), 0)
+    back = cache_to_legacy(filled)
+    assert isinstance(back, tuple) and back[0][0].shape == (1, 2, 3, 4)
+
+    # packaging follows the template type
+    assert isinstance(cache_like(list(legacy), None), tuple)
+    rebuilt = cache_like(list(legacy), DynamicCache())
+    assert isinstance(rebuilt, DynamicCache)
+    assert rebuilt.get_seq_length() == 3
+    assert cache_like([], DynamicCache()) is None
