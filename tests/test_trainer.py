@@ -384,3 +384,15 @@ def test_args_max_train_tokens():
         "--total_batch_size", "4", "--max_train_tokens", "1B",
     ])
     assert args.max_train_tokens == 1_000_000_000
+
+
+def test_regime_trainable_scaling(tmp_path):
+    """--train_scaling (tanh-parameterized lora scale, reference
+    relora.py trainable_scaling) through the full loop incl. merges."""
+    args = run_args(tmp_path, extra=["--train_scaling"], steps=4)
+    main(args)
+    state = json.load(open(tmp_path / "run" / "model_4" / "training_state.json"))
+    assert state["n_lora_restarts"] >= 1
+    sd = torch.load(tmp_path / "run" / "model_4" / "pytorch_model.bin",
+                    map_location="cpu", weights_only=True)
+    assert any("scaling" in k for k in sd), list(sd)[:5]
