@@ -389,10 +389,10 @@ def test_args_max_train_tokens():
 def test_regime_trainable_scaling(tmp_path):
     """--train_scaling (tanh-parameterized lora scale, reference
     relora.py trainable_scaling) through the full loop incl. merges."""
-    args = run_args(tmp_path, extra=["--train_scaling"], steps=4)
+    args = run_args(tmp_path, extra=["--train_scaling"], steps=6)
     main(args)
-    state = json.load(open(tmp_path / "run" / "model_4" / "training_state.json"))
+    state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
     assert state["n_lora_restarts"] >= 1
-    sd = torch.load(tmp_path / "run" / "model_4" / "pytorch_model.bin",
+    sd = torch.load(tmp_path / "run" / "model_6" / "pytorch_model.bin",
                     map_location="cpu", weights_only=True)
     assert any("scaling" in k for k in sd), list(sd)[:5]
