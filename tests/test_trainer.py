@@ -396,3 +396,11 @@ def test_regime_trainable_scaling(tmp_path):
     sd = torch.load(tmp_path / "run" / "model_6" / "pytorch_model.bin",
                     map_location="cpu", weights_only=True)
     assert any("scaling" in k for k in sd), list(sd)[:5]
+
+
+def test_regime_quantized_int8(tmp_path):
+    """--quantize 8bit end-to-end on CPU (int8 blockwise frozen weights)."""
+    args = run_args(tmp_path, extra=["--quantize", "8bit"])
+    main(args)
+    state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
+    assert state["update_step"] == 6 and state["n_lora_restarts"] >= 1
