@@ -404,3 +404,22 @@ def test_regime_quantized_int8(tmp_path):
     main(args)
     state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
     assert state["update_step"] == 6 and state["n_lora_restarts"] >= 1
+
+
+def test_regime_combo_quantized_zero_magprune_resume(tmp_path):
+    """The heaviest interaction: NF4 frozen weights + ZeRO-1 + magnitude
+    pruning + interrupt/autoresume, end-to-end (a combination matrix the
+    reference never exercised)."""
+    extra = ["--quantize", "4bit", "--optimizer", "adam_zero",
+             "--reset_optimizer_on_relora", "False",
+             "--optimizer_magnitude_pruning", "0.9",
+             "--save_every", "3"]
+    import torch.distributed as dist
+
+    main(run_args(tmp_path, extra=extra, steps=3))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    main(run_args(tmp_path, extra=extra + ["--autoresume", "true"], steps=6))
+    state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
+    assert state["update_step"] == 6
+    assert state["n_optimizer_resets"] >= 1
