@@ -243,3 +243,52 @@ def test_lora_rowmajor_v2_swizzle_conflict_free_at_r_multiple_64():
             for addrs in gen:
                 c, m = access_cycles(addrs, kind)
                 assert c == m, (r, kind, c, m)
+
+
+def test_v2_dominates_current_at_all_rtiles():
+    """skinny_grad stages rtile=min(r-r0,128) rows; Q_V2's conflict profile
+    is better-or-equal to the current rotation at every rtile: fragment
+    reads conflict-free everywhere (currently 2-way), staging writes at the
+    same level."""
+    from lds_bank_model import access_cycles, tr64, tr64_v2
+
+    def worst(gen, kind):
+        w = 1
+        for addrs in gen:
+            c, m = access_cycles(addrs, kind)
+            w = max(w, c // m if m else 1)
+        return w
+
+    for rtile in (32, 64, 96, 128):
+        def pt_writes(fn):
+            total = 64 * (rtile // 8)
+            for it in range((total + 255) // 256):
+                for w in range(4):
+                    for j in range(8):
+                        addrs = {}
+                        for lane in range(64):
+                            t = w * 64 + lane + it * 256
+                            if t >= total:
+                                continue
+                            mm = t // (rtile // 8)
+                            j8 = (t % (rtile // 8)) * 8
+                            addrs[lane] = 2 * fn(j8 + j, mm)
+                        if addrs:
+                            yield addrs
+
+        def a_reads(fn):
+            for w in range(4):
+                for i in range(2):
+                    for ks in (0, 1):
+                        addrs = {}
+                        for l in range(64):
+                            jrow = w * 32 + i * 16 + (l & 15)
+                            if jrow >= rtile:
+                                continue
+                            addrs[l] = 2 * fn(jrow, ks * 32 + (l >> 4) * 8)
+                        if addrs:
+                            yield addrs
+
+        assert worst(a_reads(tr64_v2), "read_b128") == 1, rtile
+        assert (worst(pt_writes(tr64_v2), "write_b16")
+                <= worst(pt_writes(tr64), "write_b16")), rtile
