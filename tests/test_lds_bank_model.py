@@ -354,3 +354,49 @@ def test_v2_transq_never_worse_than_current():
     # r=64 reads fully conflict-free under v2
     _, tr_v64 = patterns(rot8_v2, 0, 64)
     assert worst(tr_v64(), "read_b128") == 1
+
+
+def test_16deep_transq_layout_reads_conflict_free():
+    """The closed-form 16-deep permutation solves the one pattern Q_V2
+    cannot (TRANSQ q_im at r=128, 256-byte rows): reads conflict-free,
+    writes at the 2-way floor, bijective."""
+    from lds_bank_model import access_cycles, perm16, transq_elem_16deep
+
+    r = 128
+
+    def tw():
+        total = r * 16
+        for it in range((total + 255) // 256):
+            for w in range(4):
+                for j in range(8):
+                    addrs = {}
+                    for lane in range(64):
+                        t = w * 64 + lane + it * 256
+                        if t >= total:
+                            continue
+                        k, nb = t // 16, (t % 16) * 8
+                        addrs[lane] = 2 * transq_elem_16deep(nb + j, k)
+                    if addrs:
+                        yield addrs
+
+    def tr():
+        for w in range(4):
+            wc = (w & 1) * 64
+            for ni in range(4):
+                for kk in range(0, r, 32):
+                    yield {l: 2 * transq_elem_16deep(wc + ni * 16 + (l & 15),
+                                                    kk + (l >> 4) * 8)
+                           for l in range(64)}
+
+    for addrs in tr():
+        c, m = access_cycles(addrs, "read_b128")
+        assert c == m, (c, m)
+    worst = 1
+    for addrs in tw():
+        c, m = access_cycles(addrs, "write_b16")
+        worst = max(worst, c // m)
+    assert worst == 2
+    assert len({transq_elem_16deep(row, k)
+                for row in range(128) for k in range(128)}) == 128 * 128
+    for row in (0, 5, 127):
+        assert len({perm16(row, g) for g in range(16)}) == 16

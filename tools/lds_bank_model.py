@@ -302,3 +302,32 @@ def attn_p_tile_reads(lpad, swz):
         yield {l: 2 * (((l & 15) * ld + ks * 32 + (l >> 4) * 8)
                        ^ rm_swz(l & 15, ld, swz))
                for l in range(64)}
+
+
+# ---------------------------------------------------------------------------
+# 16-deep whole-row permutation for 256-byte-stride transposed tiles (the
+# lora_add_nn TRANSQ q_im at r=128): any 8-deep rotation is pigeonhole-bound
+# to 2-way reads there (slot%16 sees only the rotation, and an 8-value
+# rotation cannot cover 16 slots).  Closed-form construction, verified
+# conflict-free reads + floor (2-way) writes by the tests:
+#   perm16(row, grp) = (X16[row&15] + 8*(grp&1) + (grp>>1) + 2*(row>>4)) % 16
+# X16 puts evens on the b128 lane-group A-columns and odds on B — the +8 of
+# a pair's second column then completes Z16 in every read group, and the
+# even/odd split keeps the j/8+j write pairs on distinct banks.
+# ---------------------------------------------------------------------------
+
+_A_COLS = (0, 1, 2, 3, 12, 13, 14, 15)
+X16 = [0] * 16
+for _i, _c in enumerate(_A_COLS):
+    X16[_c] = 2 * _i
+for _c in range(4, 12):
+    X16[_c] = 2 * (_c - 4) + 1
+
+
+def perm16(row, grp):
+    return (X16[row & 15] + 8 * (grp & 1) + (grp >> 1) + 2 * (row >> 4)) % 16
+
+
+def transq_elem_16deep(row, k, ldt=128):
+    """TRANSQ tile element for 256-byte-stride rows (r=128)."""
+    return row * ldt + perm16(row, (k & 127) >> 3) * 8 + (k & 7)
