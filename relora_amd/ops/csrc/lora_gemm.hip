@@ -78,6 +78,34 @@ DEV_INLINE int rot8(int row, int c64) {
 #endif
 DEV_INLINE int tr64(int row, int c) { return row * 64 + rot8(row, c); }
 
+// In-row offset for a TRANSQ (transposed) tile element at column k of a row
+// with r k-elements.  Default: 64-element blocks with the 8-deep rot8
+// rotation.  Under RELORA_AMD_ROT_V2 with r a multiple of 128 (256-byte
+// rows), an 8-deep rotation is pigeonhole-bound to 2-way fragment reads,
+// so use the closed-form 16-deep whole-row permutation instead (verified
+// conflict-free reads + floor writes by the in-tree bank model).  The
+// writer and reader share this one mapping, so the choice is
+// numerics-neutral.
+#ifdef RELORA_AMD_ROT_V2
+__device__ constexpr unsigned char X16_TRQ[16] = {
+    0, 2, 4, 6, 1, 3, 5, 7, 9, 11, 13, 15, 8, 10, 12, 14};
+DEV_INLINE int trq_off(int row, int k, int r) {
+  if ((r & 127) == 0) {
+    const int grp = (k & 127) >> 3;
+    const int p = (X16_TRQ[row & 15] + ((grp & 1) << 3) + (grp >> 1) +
+                   2 * (row >> 4)) & 15;
+    return (k & ~127) + p * 8 + (k & 7);
+  }
+  return (k & ~63) + rot8(row, k & 63);
+}
+#else
+DEV_INLINE int trq_off(int row, int k, int r) {
+  (void)r;
+  return (k & ~63) + rot8(row, k & 63);
+}
+#endif
+
+
 // ---------------------------------------------------------------------------
 // philox4x32-10 — counter-based RNG for the dropout mask (regenerable, but we
 // persist packed bits: exact replay in backward with zero recompute).
@@ -201,7 +229,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        q_im[(nb + j) * ldt + (k & ~63) + rot8(nb + j, k & 63)] = v[j];
+        q_im[(nb + j) * ldt + trq_off(nb + j, k, r)] = v[j];
     }
   }
   __syncthreads();
@@ -229,7 +257,7 @@ __global__ __launch_bounds__(256) void lora_skinny_kernel(
         const int qrow = wc + ni * 16 + fr;
         const int koff = kk + kg;
         const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            TRANSQ ? q_im + qrow * ldt + (koff & ~63) + rot8(qrow, koff & 63)
+            TRANSQ ? q_im + qrow * ldt + trq_off(qrow, koff, r)
                    : q_im + rm_idx(qrow, koff, ldt));
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[mi][ni], 0, 0, 0);
       }
