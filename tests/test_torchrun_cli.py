@@ -9,6 +9,8 @@ import sys
 
 import pytest
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
 
 @pytest.mark.timeout(600)
 def test_torchrun_two_proc_cli(tmp_path):
@@ -117,3 +119,18 @@ def test_torchrun_megatron_zero_two_proc(tmp_path):
     opt = __import__("torch").load(tmp_path / "run" / "model_4" / "optimizer.pt",
                                    map_location="cpu", weights_only=False)
     assert "optimizer" in opt and opt["optimizer"]["state"]
+
+
+@pytest.mark.parametrize("script", ["torchrun_main.py", "run_glue.py",
+                                    "pretokenize.py", "bench.py"])
+def test_cli_help_exits_zero(script):
+    """Every entry-point script imports cleanly and prints --help (guards
+    import-time breakage of the CLI surface)."""
+    import subprocess
+    import sys as _sys
+
+    r = subprocess.run([_sys.executable, os.path.join(REPO, script), "--help"],
+                       capture_output=True, text=True, timeout=180,
+                       env={**os.environ, "PYTHONPATH": REPO}, cwd=REPO)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "usage" in r.stdout.lower() or "usage" in r.stderr.lower()
