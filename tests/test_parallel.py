@@ -324,3 +324,75 @@ def test_ddp_zero1_world4():
         p.join(timeout=60)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+def _zero_reset_worker(rank, world, port, q):
+    """ReLoRA's optimizer_reset against a REAL sharded ZeRO optimizer
+    (2 processes): pruning must hit the rank-local shard states through the
+    `.optim.state` surface (the reference's documented ZeRO quirk,
+    training_utils.py:267-364) and training must continue."""
+    try:
+        _init(rank, world, port)
+        from relora_amd import training_utils
+        from relora_amd.parallel import DistributedModel, ZeroRedundancyAdamW
+
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(torch.nn.Linear(16, 32), torch.nn.Linear(32, 8))
+        wrapped = DistributedModel(model)
+        params = [p for p in model.parameters() if p.requires_grad]
+        opt = ZeroRedundancyAdamW(params, lr=1e-2, betas=(0.9, 0.999))
+
+        torch.manual_seed(5)
+        for _ in range(3):
+            x = torch.randn(4, 16)
+            wrapped.set_gradient_sync(True)
+            wrapped(x).pow(2).mean().backward()
+            wrapped.finish_gradient_sync()
+            opt.step()
+            wrapped.zero_grad_buffers()
+
+        # states are dense before reset on the owning rank
+        owned = [p for p in params if p in opt.optim.state
+                 and "exp_avg" in opt.optim.state[p]]
+        assert owned, "rank owns no shard states"
+        pre_zero = [float((opt.optim.state[p]["exp_avg"] == 0).float().mean())
+                    for p in owned]
+
+        training_utils.optimizer_reset(
+            opt, reset_params=params,
+            optimizer_state_keys=["exp_avg", "exp_avg_sq"],
+            reset_optimizer_on_relora=False,
+            optimizer_random_pruning=0.0,
+            optimizer_magnitude_pruning=0.9,
+        )
+        post_zero = [float((opt.optim.state[p]["exp_avg"] == 0).float().mean())
+                     for p in owned]
+        ok = all(b > a and b >= 0.85 for a, b in zip(pre_zero, post_zero))
+
+        # loop keeps running after the reset
+        x = torch.randn(4, 16)
+        wrapped.set_gradient_sync(True)
+        wrapped(x).pow(2).mean().backward()
+        wrapped.finish_gradient_sync()
+        opt.step()
+        q.put((rank, ok, None if ok else (pre_zero, post_zero)))
+    except Exception as e:  # pragma: no cover
+        q.put((rank, False, repr(e)))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_zero_optimizer_reset_on_shards():
+    world, port = 2, free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_zero_reset_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, ok, err in results:
+        assert ok, f"rank {rank}: {err}"
