@@ -423,3 +423,36 @@ def test_regime_combo_quantized_zero_magprune_resume(tmp_path):
     state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
     assert state["update_step"] == 6
     assert state["n_optimizer_resets"] >= 1
+
+
+def test_nan_consensus_batch_skip(tmp_path, monkeypatch):
+    """NaN losses skip the optimizer update (C3 consensus path, reference
+    torchrun_main.py:813-822) but training continues and completes."""
+    from relora_amd.models import llama as llama_mod
+
+    orig_forward = llama_mod.LlamaForCausalLM.forward
+    calls = {"n": 0}
+
+    def nan_every_third(self, *a, **k):
+        out = orig_forward(self, *a, **k)
+        if self.training and out.loss is not None:
+            calls["n"] += 1
+            if calls["n"] % 3 == 0:
+                out.loss = out.loss * float("nan")
+        return out
+
+    monkeypatch.setattr(llama_mod.LlamaForCausalLM, "forward", nan_every_third)
+    # clipping must be off: with error_if_nonfinite clipping, NaN grads crash
+    # loudly BEFORE the consensus (reference semantics, torchrun_main.py:806)
+    args = run_args(tmp_path, extra=["--clip_grad_norm", "0"], steps=6)
+    main(args)
+    # with 1/3 of batches NaN, the 5%-skipped abort fires (reference
+    # torchrun_main.py:819-822) and the final save happens at the abort step
+    ckpts = sorted((tmp_path / "run").glob("model_*"))
+    assert ckpts, "no final checkpoint written"
+    state = json.load(open(ckpts[-1] / "training_state.json"))
+    assert state["update_step"] <= 6
+    sd = torch.load(ckpts[-1] / "pytorch_model.bin",
+                    map_location="cpu", weights_only=True)
+    for k, v in sd.items():
+        assert torch.isfinite(v).all(), k
