@@ -456,3 +456,21 @@ def test_nan_consensus_batch_skip(tmp_path, monkeypatch):
                     map_location="cpu", weights_only=True)
     for k, v in sd.items():
         assert torch.isfinite(v).all(), k
+
+
+def test_eval_nan_raises(tmp_path, monkeypatch):
+    """A NaN during evaluation raises (reference evaluate_model,
+    torchrun_main.py:176-178) rather than silently logging."""
+    from relora_amd.models import llama as llama_mod
+
+    orig_forward = llama_mod.LlamaForCausalLM.forward
+
+    def nan_in_eval(self, *a, **k):
+        out = orig_forward(self, *a, **k)
+        if not self.training and out.loss is not None:
+            out.loss = out.loss * float("nan")
+        return out
+
+    monkeypatch.setattr(llama_mod.LlamaForCausalLM, "forward", nan_in_eval)
+    with pytest.raises(RuntimeError, match="nan"):
+        main(run_args(tmp_path, steps=3))
