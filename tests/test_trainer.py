@@ -474,3 +474,21 @@ def test_eval_nan_raises(tmp_path, monkeypatch):
     monkeypatch.setattr(llama_mod.LlamaForCausalLM, "forward", nan_in_eval)
     with pytest.raises(RuntimeError, match="nan"):
         main(run_args(tmp_path, steps=3))
+
+
+def test_skip_batches_runtime(tmp_path):
+    """--skip_batches skips the named update steps (no loss logged for
+    them) while still reaching num_training_steps (reference
+    torchrun_main.py:772-775)."""
+    args = run_args(tmp_path, extra=["--skip_batches", "2"], steps=6)
+    main(args)
+    state = json.load(open(tmp_path / "run" / "model_6" / "training_state.json"))
+    assert state["update_step"] == 6
+    logged = [json.loads(l) for l in
+              open(tmp_path / "run" / "wandb_offline.jsonl")]
+    steps_with_loss = {r["update_step"] for r in logged if "loss" in r}
+    # the skip matches the PRE-increment update_step (reference semantics:
+    # "update_step numbers", torchrun_main.py:772), so skipping "2"
+    # suppresses the update that would log as step 3
+    assert 3 not in steps_with_loss
+    assert {1, 2, 4, 5, 6} <= steps_with_loss
