@@ -492,3 +492,14 @@ def test_skip_batches_runtime(tmp_path):
     # suppresses the update that would log as step 3
     assert 3 not in steps_with_loss
     assert {1, 2, 4, 5, 6} <= steps_with_loss
+
+
+def test_mid_run_eval_fires(tmp_path):
+    """--eval_every triggers mid-run evals, logged to the recorder
+    (reference torchrun_main.py eval window)."""
+    args = run_args(tmp_path, extra=["--eval_every", "3"], steps=6)
+    main(args)
+    logged = [json.loads(l) for l in
+              open(tmp_path / "run" / "wandb_offline.jsonl")]
+    evals = [r for r in logged if "final_eval_loss" in r]
+    assert len(evals) >= 2  # at steps 3 and 6
