@@ -14,7 +14,10 @@ def cache_to_legacy(past):
     # a freshly-created DynamicCache may hold initialized-but-empty layers
     if len(past) == 0 or past.get_seq_length() == 0:
         return None
-    return tuple((layer.keys, layer.values) for layer in past.layers)
+    if hasattr(past, "layers"):
+        return tuple((layer.keys, layer.values) for layer in past.layers)
+    # older transformers: DynamicCache exposes key_cache/value_cache lists
+    return tuple(zip(past.key_cache, past.value_cache))
 
 
 def cache_like(next_cache, template) -> Optional[object]:

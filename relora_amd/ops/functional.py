@@ -9,9 +9,13 @@ Each op has two implementations:
 
 Kernel inventory parity: K1/K2 `lora_linear` (fused LoRA
 GEMM), K3 `flash_attention`, K4 `rope`, K5 `rmsnorm`, K6 `layernorm`,
-K7 `swiglu`, K10 `fused_cross_entropy` (chunked; never materializes the
-[M,V] logits — the reference flags this memory hot spot at
-modeling_llama.py:696-697), K11/K12 live in :mod:`relora_amd.ops.optim`,
+K7 `swiglu`, K10 `fused_cross_entropy` (chunked; never materializes *fp32*
+logits — row stats and the in-place grad are fp32-accurate HIP kernels over
+bf16 logit chunks. The default single-chunk config does hold one full
+[M,V] bf16 logits buffer (~1.6 GB at the flagship shape — cheap against
+288 GB HBM); set RELORA_AMD_CE_CHUNK lower to bound that. The reference
+flags this memory hot spot at modeling_llama.py:696-697),
+K11/K12 live in :mod:`relora_amd.ops.optim`,
 K13 `merge_and_reinit` in :mod:`relora_amd.relora`, K14 pruning in
 :mod:`relora_amd.training_utils`.
 """
@@ -233,9 +237,11 @@ def flash_attention(q, k, v, causal=True, dropout_p=0.0, scale=None):
 
 # ---------------------------------------------------------------------------
 # Fused chunked cross-entropy over the LM head (K10).
-# loss = CE(shift(hidden @ Wᵀ), shift(labels)) without ever materializing the
-# full [M, V] logits: per-M-chunk GEMM + one-pass row stats (HIP) and an
-# in-place softmax-minus-onehot gradient kernel in backward.
+# loss = CE(shift(hidden @ Wᵀ), shift(labels)) without materializing fp32
+# logits: per-M-chunk bf16 GEMM + one-pass fp32 row stats (HIP) and an
+# in-place softmax-minus-onehot gradient kernel in backward.  Peak logits
+# memory is one [chunk, V] bf16 buffer (the full [M, V] only in the default
+# single-chunk config, sized for 288 GB HBM — see _CE_CHUNK below).
 # ---------------------------------------------------------------------------
 
 # 288 GB HBM comfortably holds one [M,V] bf16 logits buffer for the flagship

@@ -82,7 +82,7 @@ class ReLoRaModel(torch.nn.Module):
         use_double_quant=False,
     ):
         if r <= 0:
-            raise ValueError("r must be positive. If you want r == 0, use the original model.")
+            raise ValueError("LoRA rank r must be >= 1; for a rank-0 (plain) layer just skip the ReLoRA wrap.")
         super().__init__()
         self.wrapped_model: nn.Module = model
         self.r = r
@@ -217,7 +217,7 @@ class ReLoRaLinear(nn.Module):
     ):
         super().__init__()
         if r <= 0:
-            raise ValueError("r must be positive. If you want r == 0, use the original model.")
+            raise ValueError("LoRA rank r must be >= 1; for a rank-0 (plain) layer just skip the ReLoRA wrap.")
         if quantize not in (None, "4bit", "8bit"):
             raise ValueError(f"quantize must be None, '4bit' or '8bit', got {quantize!r}")
 

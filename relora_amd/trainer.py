@@ -599,6 +599,12 @@ def main(args):
             update_step = optimizer_checkpoint["update_step"]
             global_step = optimizer_checkpoint["global_step"]
             _rng = optimizer_checkpoint.get("rng_states_per_rank")
+            if _rng and len(_rng) != world_size:
+                logger.warning(
+                    f"Checkpoint holds RNG streams for {len(_rng)} ranks but "
+                    f"world_size is {world_size}; bit-exact resume is only "
+                    "guaranteed at the original world size (extra ranks keep "
+                    "fresh seeds / streams are reused by position).")
             if _rng and global_rank < len(_rng):
                 _st = _rng[global_rank]
                 torch.set_rng_state(_st["torch"])

@@ -10,8 +10,10 @@ random-prune standing in for a hard zero), `get_last_training_state`
 (:248-264), `delete_old_checkpoints` (:406-418),
 `max_train_tokens_to_number` (:239-245).
 
-On GPU, magnitude pruning's quantile threshold and the masked multiply run
-as HIP kernels (ops/csrc); the torch path is the CPU/test oracle.
+Pruning runs as torch ops (quantile + masked multiply) on whatever device
+the optimizer state lives on; it fires once per ReLoRA cycle (every few
+thousand steps) and is far off the hot path, so it has no dedicated HIP
+kernel.
 """
 
 import json

@@ -29,7 +29,7 @@ _LEGACY_BUFFER_KEYS = ("rotary_emb.inv_freq", "attention.bias",
 
 def _drop_legacy_buffers(sd):
     return {k: v for k, v in sd.items()
-            if not any(k.endswith(s) or s in k for s in _LEGACY_BUFFER_KEYS)}
+            if not any(k.endswith(s) for s in _LEGACY_BUFFER_KEYS)}
 
 
 def load_state_dict_compat(path):

@@ -63,7 +63,10 @@ def auto_micro_batch(cfg, seq_len, total_batch_size, world_size,
     budget = hbm_bytes * _SAFETY
     per_rank_cap = max(1, total_batch_size // world_size)
     bs = 1
-    while bs * 2 <= per_rank_cap:
+    # only grow while the result still divides the per-rank batch, so the
+    # trainer's `grad_accum * bs * world == total_batch_size` algebra holds
+    # (e.g. per_rank_cap=24 stops at 8, not 16)
+    while bs * 2 <= per_rank_cap and per_rank_cap % (bs * 2) == 0:
         need = estimate_step_bytes(cfg, bs * 2, seq_len, dtype,
                                    trainable_ratio, lora_r)
         if need > budget:

@@ -338,6 +338,8 @@ def sdpa_ref_fp32(q, k, v, causal=True):
     (1, 2, 2048, 128),    # llama_7b head
     (2, 2, 2048, 64),     # llama_1b head
     (1, 1, 96, 32),
+    (1, 2, 256, 80),      # HD=96 template, padded (RF=2 fwd) — first hw exercise
+    (1, 2, 256, 96),      # HD=96 template, exact
 ])
 def test_attn_fwd(B, nh, S, hd):
     torch.manual_seed(0)
@@ -358,6 +360,8 @@ def test_attn_fwd(B, nh, S, hd):
     (1, 2, 333, 48),
     (1, 1, 512, 128),
     (2, 2, 256, 64),
+    (1, 2, 256, 80),      # HD=96 template, padded
+    (1, 2, 256, 96),      # HD=96 template, exact
 ])
 def test_attn_bwd(B, nh, S, hd):
     torch.manual_seed(0)
