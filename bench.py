@@ -37,6 +37,10 @@ def parse_args():
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--full_rank", action="store_true",
                    help="bench full-rank training instead of ReLoRA")
+    p.add_argument("--suite", action="store_true",
+                   help="also bench the secondary configs (llama_250m r=128 "
+                        "bs8, llama_7b r=256 bs4) before the flagship; one "
+                        "JSON line each, flagship printed LAST")
     return p.parse_args()
 
 
@@ -55,7 +59,11 @@ def main():
         device = "cpu"
         backend = "gloo"
 
-    if world_size > 1 and not dist.is_initialized():
+    # init the process group whenever launched under torchrun (RANK set),
+    # including world_size 1: that makes a single-GPU `torchrun
+    # --nproc-per-node 1` run exercise RCCL init + the bucketed all-reduce
+    # path for real (with RELORA_AMD_FORCE_SYNC=1).
+    if (world_size > 1 or "RANK" in os.environ) and not dist.is_initialized():
         dist.init_process_group(backend=backend, rank=rank, world_size=world_size)
 
     from relora_amd.models import build_model_from_config, load_model_config
@@ -85,11 +93,17 @@ def main():
     optimizer = AdamW(trainable, lr=4e-4, betas=(0.9, 0.95), weight_decay=0.01)
 
     B, S = args.batch_size, args.seq_len
+    # fresh synthetic batch per step, pre-generated on device OUTSIDE the
+    # timed region, so throughput cannot benefit from single-batch caching
+    # effects and loss stays a sanity signal rather than memorization
     g = torch.Generator(device="cpu").manual_seed(42 + rank)
-    batch = torch.randint(0, cfg.vocab_size, (B, S), generator=g).to(device)
+    batches = [
+        torch.randint(0, cfg.vocab_size, (B, S), generator=g).to(device)
+        for _ in range(args.warmup + args.steps)
+    ]
     loss_info = torch.zeros(3, device=device)
 
-    def step():
+    def step(batch):
         model.set_gradient_sync(True)
         loss = model(input_ids=batch, labels=batch).loss
         loss_info[0] = loss.detach()
@@ -104,21 +118,33 @@ def main():
         model.zero_grad_buffers()
         return loss
 
-    for _ in range(args.warmup):
-        step()
+    for i in range(args.warmup):
+        step(batches[i])
+
+    # per-step boundaries via CUDA events (no per-step host sync inside the
+    # timed region); the headline time stays the single host-clock bracket
+    ev = [torch.cuda.Event(enable_timing=True)
+          for _ in range(args.steps + 1)] if use_gpu else None
 
     if dist.is_initialized():
         dist.barrier()
     if use_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        loss = step()
+    for i in range(args.steps):
+        if ev:
+            ev[i].record()
+        loss = step(batches[args.warmup + i])
+    if ev:
+        ev[args.steps].record()
     if use_gpu:
         torch.cuda.synchronize()
     if dist.is_initialized():
         dist.barrier()
     elapsed = time.perf_counter() - t0
+
+    step_ms = ([ev[i].elapsed_time(ev[i + 1]) for i in range(args.steps)]
+               if ev else [elapsed / args.steps * 1000] * args.steps)
 
     # max over ranks
     if dist.is_initialized():
@@ -155,6 +181,11 @@ def main():
                 "trainable_params_M": round(n_trainable / 1e6, 1),
                 "percent_trainable": round(100 * n_trainable / n_total, 1),
                 "final_loss": round(float(loss.detach()), 4),
+                "step_ms_min": round(min(step_ms), 2),
+                "step_ms_max": round(max(step_ms), 2),
+                "step_ms_stdev": round(
+                    (sum((t - sum(step_ms) / len(step_ms)) ** 2
+                         for t in step_ms) / len(step_ms)) ** 0.5, 2),
             },
         }
         print(json.dumps(result))
@@ -163,5 +194,24 @@ def main():
         dist.destroy_process_group()
 
 
-if __name__ == "__main__":
+def _run_suite():
+    """Secondary configs as subprocesses (fresh HIP context each), flagship
+    last so last-line JSON parsing still lands on the headline number."""
+    import subprocess
+    import sys
+    base = [a for a in sys.argv[1:] if a != "--suite"]
+    for extra in (
+        ["--model", "configs/llama_250m.json", "--batch_size", "8"],
+        ["--model", "configs/llama_7b.json", "--batch_size", "4",
+         "--lora_r", "256"],
+    ):
+        subprocess.run([sys.executable, __file__] + base + extra, check=False)
     main()
+
+
+if __name__ == "__main__":
+    import sys
+    if "--suite" in sys.argv:
+        _run_suite()
+    else:
+        main()

@@ -33,7 +33,14 @@ import torch.nn as nn
 
 
 def _dist_active():
-    return dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1
+    # RELORA_AMD_FORCE_SYNC=1 runs the collectives even at world_size 1
+    # (RCCL executes 1-rank all-reduce as a copy) so the RCCL code path —
+    # init, AVG-in-collective, async bucket overlap — can be exercised and
+    # profiled on a single GPU.
+    if not (dist.is_available() and dist.is_initialized()):
+        return False
+    return dist.get_world_size() > 1 or \
+        os.environ.get("RELORA_AMD_FORCE_SYNC", "0") == "1"
 
 
 class _Bucket:
@@ -83,6 +90,11 @@ class DistributedModel(nn.Module):
                         continue
                     dist.broadcast(t.data, src=0, group=self.process_group)
 
+        # comm accounting (visible proof the RCCL path ran: count + bytes)
+        self._comm_calls = 0
+        self._comm_bytes = 0
+        self._comm_logged = False
+
         self._buckets: List[_Bucket] = []
         self._param_bucket = {}
         self._build_buckets()
@@ -119,15 +131,21 @@ class DistributedModel(nn.Module):
             b.handle = None
 
     # -- hooks --------------------------------------------------------------
+    def _launch_reduce(self, bucket):
+        self._comm_calls += 1
+        self._comm_bytes += bucket.buffer.numel() * bucket.buffer.element_size()
+        return dist.all_reduce(
+            bucket.buffer, op=self._reduce_op, group=self.process_group,
+            async_op=True,
+        )
+
     def _on_grad_ready(self, param):
         if not self.require_backward_grad_sync or not _dist_active():
             return
         b = self._param_bucket[param]
         b.pending -= 1
         if b.pending == 0:
-            b.handle = dist.all_reduce(
-                b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
-            )
+            b.handle = self._launch_reduce(b)
 
     # -- public API ---------------------------------------------------------
     def forward(self, *args, **kwargs):
@@ -144,17 +162,10 @@ class DistributedModel(nn.Module):
             self._reset_pending()
             return
         for b in self._buckets:
-            if b.handle is None and self.require_backward_grad_sync and b.pending > 0 \
-                    and b.pending < len(b.params):
-                # partial bucket (should not happen in normal training)
-                b.handle = dist.all_reduce(
-                    b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
-                )
             if b.handle is None and self.require_backward_grad_sync:
-                # bucket never fired (e.g. unused params): reduce it so ranks agree
-                b.handle = dist.all_reduce(
-                    b.buffer, op=self._reduce_op, group=self.process_group, async_op=True
-                )
+                # bucket never fired or fired partially (e.g. unused params):
+                # reduce it so ranks agree
+                b.handle = self._launch_reduce(b)
         for b in self._buckets:
             if b.handle is not None:
                 b.handle.wait()
@@ -163,6 +174,15 @@ class DistributedModel(nn.Module):
             inv = 1.0 / self.world_size
             for b in self._buckets:
                 b.buffer.mul_(inv)
+        if not self._comm_logged and self._comm_calls:
+            self._comm_logged = True
+            from relora_amd.utils.logging import logger
+            logger.info(
+                f"grad sync active: backend={dist.get_backend(self.process_group)} "
+                f"world={dist.get_world_size(self.process_group)} "
+                f"op={'AVG' if self._reduce_op == dist.ReduceOp.AVG else 'SUM/div'} "
+                f"buckets={len(self._buckets)} "
+                f"bytes/boundary={sum(bb.buffer.numel() * bb.buffer.element_size() for bb in self._buckets) / 1e6:.1f} MB")
         self._reset_pending()
 
     def zero_grad_buffers(self):
