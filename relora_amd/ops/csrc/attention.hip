@@ -384,6 +384,232 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// forward v3 — swapped-operand 32x32x16 structure, softmax fully lane-local
+//
+// Per the CDNA4 guide's fused-attention pattern: compute S^T = K·Q^T
+// (mfma(K, Q)) so the C-fragment's *column* index — which IS the lane index
+// — is the query row.  Each lane then owns one q-row's scores (its 32 of
+// the 64-kv tile; the l^32 partner lane owns the other 32), so the online
+// softmax (max, exp, sum, rescale decision) is per-lane scalar code with a
+// single v_permlane32_swap to merge partner halves — no cross-lane shfl
+// chains and no P relayout through LDS.  P is packed to bf16 in-register
+// (pair packs + two permlane32_swaps per k-step) directly into the MFMA
+// B-fragment layout, and PV runs swapped too (O^T = V^T·P^T), which keeps
+// the q index lane-local in the O accumulator: the online rescale and the
+// final 1/l are per-lane scalar multiplies.  8 waves x 32 q-rows = 256
+// q-rows per block; 64-row K/V tiles double-buffered with the same
+// register-prefetch staging as v2.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// exchange a value with the lane^32 partner and combine
+DEV_INLINE float swap_combine_max(float x) {
+  auto r = __builtin_amdgcn_permlane32_swap(__float_as_int(x), __float_as_int(x), false, false);
+  return fmaxf(__int_as_float(r[0]), __int_as_float(r[1]));
+}
+DEV_INLINE float swap_combine_sum(float x) {
+  auto r = __builtin_amdgcn_permlane32_swap(__float_as_int(x), __float_as_int(x), false, false);
+  return __int_as_float(r[0]) + __int_as_float(r[1]);
+}
+
+DEV_INLINE unsigned pack_bf16(float lo, float hi) {
+  // scalar casts; the compiler fuses the pair (guide: hand-written
+  // v_cvt_pk_bf16_f32 asm measured slower than letting it)
+  union { __bf16 h[2]; unsigned u; } r;
+  r.h[0] = (__bf16)lo;
+  r.h[1] = (__bf16)hi;
+  return r.u;
+}
+
+// Build the P^T B-fragment for MFMA k-step s (k = kv in [16s, 16s+16)) from
+// the 32 per-lane S^T C-layout values of one kv subtile (p[0..15] = regs of
+// subtile t = s>>1).  C-layout: reg r holds kv = 32t + (r&3) + 8*(r>>2) +
+// 4*hi.  The fragment needs kv = 16s + hi*8 + [0,8): own even-quads supply
+// half, the l^32 partner the other half — one permlane32_swap per dword
+// pair completes it (guide T12).
+DEV_INLINE bf16x8 pack_p_frag(const float* p16, int g) {
+  unsigned a01 = pack_bf16(p16[8 * g + 0], p16[8 * g + 1]);
+  unsigned a23 = pack_bf16(p16[8 * g + 2], p16[8 * g + 3]);
+  unsigned b01 = pack_bf16(p16[8 * g + 4], p16[8 * g + 5]);
+  unsigned b23 = pack_bf16(p16[8 * g + 6], p16[8 * g + 7]);
+  auto r0 = __builtin_amdgcn_permlane32_swap((int)a01, (int)b01, false, false);
+  auto r1 = __builtin_amdgcn_permlane32_swap((int)a23, (int)b23, false, false);
+  union { unsigned u[4]; bf16x8 v; } f;
+  f.u[0] = (unsigned)r0[0];
+  f.u[1] = (unsigned)r1[0];
+  f.u[2] = (unsigned)r0[1];
+  f.u[3] = (unsigned)r1[1];
+  return f.v;
+}
+
+#define LOG2E 1.44269504088896340736f
+#define DEFER_MAX_THR 11.5f  // log2 domain ~ e^8 (guide T13; bf16 accum headroom)
+
+template <int HD>
+__global__ __launch_bounds__(512) void attn_fwd_v3_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
+    float* __restrict__ lse, int S, int hd, float scale) {
+  constexpr int KSTEPS = HD / 16;  // QK^T k-steps per 32-kv subtile
+  constexpr int NT32 = HD / 32;    // O^T 32-row hd tiles
+  constexpr int LDK = HD + LPAD;
+  constexpr int NV = (HD + 63) / 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds_k = (__bf16*)smem;            // [2][TILE][LDK] K rows
+  __bf16* lds_vt = lds_k + 2 * TILE * LDK;  // [2][HD][TILE]  V^T rotated
+
+  const int bh = blockIdx.y;
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * 256;  // heavy blocks first
+  const long base = (long)bh * S * hd;
+  const __hip_bfloat16* qp = q + base;
+  const __hip_bfloat16* kp = k + base;
+  const __hip_bfloat16* vp = v + base;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int qcol = lane & 31;  // this lane's q row (within the wave's 32)
+  const int hi = lane >> 5;    // k-run half selector
+  const int q_abs = q_start + wave * 32 + qcol;
+
+  // Q as B-fragments: lane holds Q[q_abs][ks*16 + hi*8 + 0..8)
+  bf16x8 qf[KSTEPS];
+#pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks)
+    qf[ks] = global_frag(qp, q_abs, S, hd, ks * 16 + hi * 8);
+
+  float m2 = -INFINITY;  // running max of scores * scale * log2e
+  float l_run = 0.f;
+  f32x16 o_acc[NT32];
+#pragma unroll
+  for (int t = 0; t < NT32; ++t) o_acc[t] = (f32x16)(0.f);
+
+  const int q_max_abs = min(q_start + 255, S - 1);
+  const int n_tiles = (q_max_abs / TILE) + 1;
+  const float sc2 = scale * LOG2E;
+
+  bf16x8 rk[NV], rv[NV];
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
+  tile_write_t<HD, NV>(lds_vt, rv);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
+  __syncthreads();
+
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int cur = kt & 1;
+    const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
+    const __bf16* vtb = lds_vt + cur * HD * TILE;
+    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
+
+    // S^T = K·Q^T over two 32-kv subtiles; lane ends with 32 score f32s
+    float p_val[2][16];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x16 acc = (f32x16)(0.f);
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks)
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            lds_frag(kb, 32 * t + qcol, ks * 16 + hi * 8, LDK), qf[ks], acc, 0, 0, 0);
+      if (edge) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          p_val[t][r] = (kv_abs > q_abs || kv_abs >= S) ? -INFINITY : acc[r] * sc2;
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) p_val[t][r] = acc[r] * sc2;
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // stage tile kt+1 while the VALU softmax runs (T14 split)
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+    }
+
+    // online softmax, fully per-lane (q = lane&31)
+    float m_tile = p_val[0][0];
+#pragma unroll
+    for (int i = 1; i < 16; ++i) m_tile = fmaxf(m_tile, p_val[0][i]);
+#pragma unroll
+    for (int i = 0; i < 16; ++i) m_tile = fmaxf(m_tile, p_val[1][i]);
+    m_tile = swap_combine_max(m_tile);
+
+    // defer-max: only rescale when the tile max exceeds the running max by
+    // more than THR (wave-uniform branch; textbook-safe order — decision
+    // precedes this tile's exponentiation, l-update follows at same scale)
+    if (__any(m_tile > m2 + DEFER_MAX_THR) || m2 == -INFINITY) {
+      const float m_new = fmaxf(m2, m_tile);
+      const float alpha = (m_new == -INFINITY) ? 0.f : exp2f(m2 - m_new);
+      l_run *= alpha;
+#pragma unroll
+      for (int t = 0; t < NT32; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[t][r] *= alpha;
+      m2 = m_new;
+    }
+
+    float rsum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float p = (m2 == -INFINITY) ? 0.f : exp2f(p_val[t][r] - m2);
+        p_val[t][r] = p;
+        rsum += p;
+      }
+    l_run += swap_combine_sum(rsum);
+
+    // PV swapped: O^T += V^T · P^T, 4 k-steps over the 64-kv tile
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 pfrag = pack_p_frag(p_val[s >> 1], s & 1);
+#pragma unroll
+      for (int ht = 0; ht < NT32; ++ht)
+        o_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ldsT_frag(vtb, ht * 32 + qcol, s * 16 + hi * 8), pfrag, o_acc[ht], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+  }
+
+  // epilogue: per-lane 1/l, packed b32 stores (hd pairs are reg pairs)
+  if (q_abs < S) {
+    const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+#pragma unroll
+    for (int ht = 0; ht < NT32; ++ht)
+#pragma unroll
+      for (int d = 0; d < 8; ++d) {
+        const int hd_c = ((2 * d) & 3) + 8 * ((2 * d) >> 2) + 4 * hi + 32 * ht;
+        if (hd_c + 1 < hd) {
+          const unsigned w =
+              pack_bf16(o_acc[ht][2 * d] * inv_l, o_acc[ht][2 * d + 1] * inv_l);
+          *reinterpret_cast<unsigned*>(out + base + (long)q_abs * hd + hd_c) = w;
+        } else if (hd_c < hd) {
+          out[base + (long)q_abs * hd + hd_c] = __float2bfloat16(o_acc[ht][2 * d] * inv_l);
+        }
+      }
+    if (hi == 0)
+      lse[(long)bh * S + q_abs] =
+          (m2 + log2f(fmaxf(l_run, 1e-30f))) * 0.6931471805599453f;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // backward: delta preprocess
 // ---------------------------------------------------------------------------
 
@@ -397,6 +623,323 @@ __global__ void attn_delta_kernel(const __hip_bfloat16* __restrict__ dout,
     acc += to_f32(dout[row * hd + i]) * to_f32(o[row * hd + i]);
   acc = block_reduce_sum(acc, scratch);
   if (threadIdx.x == 0) delta[row] = acc;
+}
+
+// ---------------------------------------------------------------------------
+// backward v3 — same swapped-operand 32x32x16 structure as the v3 forward.
+//
+// dQ kernel: S^T = K·Q^T and dP^T = V·dO^T keep the q index lane-local
+// (one lse/delta load per lane); dS^T is packed in-register into B-frags
+// and dQ^T = K^T·dS^T accumulates with q lane-local — per-lane epilogue.
+//
+// dK/dV kernel: S = Q·K^T and dP = dO·V^T keep the KV index lane-local
+// (dk/dv accumulators stay in this block); lse/delta are read as guarded
+// f32x4 quads (the C-layout row quads are q-contiguous); P^T and dS^T pack
+// in-register into A-frags for dV += P^T·dO and dK += dS^T·Q.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
+DEV_INLINE f32x4 load_f32x4_guard(const float* p, long idx, long n) {
+  f32x4 r;
+  if (idx + 3 < n) {
+    r = *reinterpret_cast<const f32x4*>(p + idx);
+  } else {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) r[i] = (idx + i < n) ? p[idx + i] : 0.f;
+  }
+  return r;
+}
+
+template <int HD>
+__global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    __hip_bfloat16* __restrict__ dq, int S, int hd, float scale) {
+  constexpr int KSTEPS = HD / 16;
+  constexpr int NT32 = HD / 32;
+  constexpr int LDK = HD + LPAD;
+  constexpr int NV = (HD + 63) / 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds_k = (__bf16*)smem;             // [2][TILE][LDK] K rows
+  __bf16* lds_v = lds_k + 2 * TILE * LDK;    // [2][TILE][LDK] V rows
+  __bf16* lds_kt = lds_v + 2 * TILE * LDK;   // [2][HD][TILE]  K^T rotated
+
+  const int bh = blockIdx.y;
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * 256;  // heavy blocks first
+  const long base = (long)bh * S * hd;
+  const __hip_bfloat16* qp = q + base;
+  const __hip_bfloat16* kp = k + base;
+  const __hip_bfloat16* vp = v + base;
+  const __hip_bfloat16* dop = dout + base;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int qcol = lane & 31;
+  const int hi = lane >> 5;
+  const int q_abs = q_start + wave * 32 + qcol;
+
+  bf16x8 qf[KSTEPS], dof[KSTEPS];
+#pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks) {
+    qf[ks] = global_frag(qp, q_abs, S, hd, ks * 16 + hi * 8);
+    dof[ks] = global_frag(dop, q_abs, S, hd, ks * 16 + hi * 8);
+  }
+  const float sc2 = scale * LOG2E;
+  const float lse2 = (q_abs < S) ? lse[(long)bh * S + q_abs] * LOG2E : 0.f;
+  const float delt = (q_abs < S) ? delta[(long)bh * S + q_abs] : 0.f;
+
+  f32x16 dq_acc[NT32];
+#pragma unroll
+  for (int t = 0; t < NT32; ++t) dq_acc[t] = (f32x16)(0.f);
+
+  const int q_max_abs = min(q_start + 255, S - 1);
+  const int n_tiles = (q_max_abs / TILE) + 1;
+
+  bf16x8 rk[NV], rv[NV];
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_write_rows<HD, NV>(lds_k, rk, LDK);
+  tile_write_rows<HD, NV>(lds_v, rv, LDK);
+  tile_write_t<HD, NV>(lds_kt, rk);
+  if (n_tiles > 1) {
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+  }
+  __syncthreads();
+
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int cur = kt & 1;
+    const int kv0 = kt * TILE;
+    const __bf16* kb = lds_k + cur * TILE * LDK;
+    const __bf16* vb = lds_v + cur * TILE * LDK;
+    const __bf16* ktb = lds_kt + cur * HD * TILE;
+    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
+
+    float ds_val[2][16];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const int c0 = ks * 16 + hi * 8;
+        sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            lds_frag(kb, 32 * t + qcol, c0, LDK), qf[ks], sa, 0, 0, 0);
+        dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            lds_frag(vb, 32 * t + qcol, c0, LDK), dof[ks], dpa, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p;
+        if (edge) {
+          const int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          p = (kv_abs <= q_abs && kv_abs < S && q_abs < S)
+                  ? exp2f(sa[r] * sc2 - lse2) : 0.f;
+        } else {
+          p = exp2f(sa[r] * sc2 - lse2);
+        }
+        ds_val[t][r] = p * (dpa[r] - delt) * scale;
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (kt + 1 < n_tiles) {
+      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
+      if (kt + 2 < n_tiles) {
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      }
+    }
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 dsf = pack_p_frag(ds_val[s >> 1], s & 1);
+#pragma unroll
+      for (int ht = 0; ht < NT32; ++ht)
+        dq_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            ldsT_frag(ktb, ht * 32 + qcol, s * 16 + hi * 8), dsf, dq_acc[ht], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+  }
+
+  if (q_abs < S) {
+#pragma unroll
+    for (int ht = 0; ht < NT32; ++ht)
+#pragma unroll
+      for (int d = 0; d < 8; ++d) {
+        const int hd_c = ((2 * d) & 3) + 8 * ((2 * d) >> 2) + 4 * hi + 32 * ht;
+        if (hd_c + 1 < hd) {
+          const unsigned w = pack_bf16(dq_acc[ht][2 * d], dq_acc[ht][2 * d + 1]);
+          *reinterpret_cast<unsigned*>(dq + base + (long)q_abs * hd + hd_c) = w;
+        } else if (hd_c < hd) {
+          dq[base + (long)q_abs * hd + hd_c] = __float2bfloat16(dq_acc[ht][2 * d]);
+        }
+      }
+  }
+}
+
+template <int HD>  // HD <= 64 (register budget); hd128 uses the v2 kernel
+__global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
+    int S, int hd, float scale) {
+  constexpr int KSTEPS = HD / 16;
+  constexpr int NT32 = HD / 32;
+  constexpr int LDK = HD + LPAD;
+  constexpr int NV = (HD + 63) / 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds_q = (__bf16*)smem;              // [2][TILE][LDK] Q rows
+  __bf16* lds_do = lds_q + 2 * TILE * LDK;    // [2][TILE][LDK] dO rows
+  __bf16* lds_qt = lds_do + 2 * TILE * LDK;   // [2][HD][TILE]  Q^T rotated
+  __bf16* lds_dot = lds_qt + 2 * HD * TILE;   // [2][HD][TILE]  dO^T rotated
+
+  const int bh = blockIdx.y;
+  const int kv_start = blockIdx.x * 256;
+  const long base = (long)bh * S * hd;
+  const __hip_bfloat16* qp = q + base;
+  const __hip_bfloat16* kp = k + base;
+  const __hip_bfloat16* vp = v + base;
+  const __hip_bfloat16* dop = dout + base;
+  const float* lsep = lse + (long)bh * S;
+  const float* deltap = delta + (long)bh * S;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int kvcol = lane & 31;
+  const int hi = lane >> 5;
+  const int kv_abs = kv_start + wave * 32 + kvcol;  // this lane's kv row
+
+  bf16x8 kf[KSTEPS], vf[KSTEPS];
+#pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks) {
+    kf[ks] = global_frag(kp, kv_abs, S, hd, ks * 16 + hi * 8);
+    vf[ks] = global_frag(vp, kv_abs, S, hd, ks * 16 + hi * 8);
+  }
+  const float sc2 = scale * LOG2E;
+
+  f32x16 dk_acc[NT32], dv_acc[NT32];
+#pragma unroll
+  for (int t = 0; t < NT32; ++t) {
+    dk_acc[t] = (f32x16)(0.f);
+    dv_acc[t] = (f32x16)(0.f);
+  }
+
+  const int first_qt = kv_start / TILE;
+  const int n_q_tiles = (S + TILE - 1) / TILE;
+
+  bf16x8 rq[NV], rdo[NV];
+  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd);
+  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
+  tile_write_rows<HD, NV>(lds_q, rq, LDK);
+  tile_write_rows<HD, NV>(lds_do, rdo, LDK);
+  tile_write_t<HD, NV>(lds_qt, rq);
+  tile_write_t<HD, NV>(lds_dot, rdo);
+  if (first_qt + 1 < n_q_tiles) {
+    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd);
+    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd);
+  }
+  __syncthreads();
+
+  for (int qt = first_qt; qt < n_q_tiles; ++qt) {
+    const int cur = qt & 1;
+    const int q0 = qt * TILE;
+    const __bf16* qb = lds_q + cur * TILE * LDK;
+    const __bf16* dob = lds_do + cur * TILE * LDK;
+    const __bf16* qtb = lds_qt + cur * HD * TILE;
+    const __bf16* dotb = lds_dot + cur * HD * TILE;
+    const bool edge = (q0 < kv_start + 255) || (q0 + TILE > S);
+
+    float pt_val[2][16], dst_val[2][16];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const int c0 = ks * 16 + hi * 8;
+        sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            lds_frag(qb, 32 * t + (lane & 31), c0, LDK), kf[ks], sa, 0, 0, 0);
+        dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            lds_frag(dob, 32 * t + (lane & 31), c0, LDK), vf[ks], dpa, 0, 0, 0);
+      }
+      // lse/delta for the C-layout q rows: quads are q-contiguous
+      f32x4 lse4[4], dl4[4];
+#pragma unroll
+      for (int qd = 0; qd < 4; ++qd) {
+        const long qrow = q0 + 32 * t + 8 * qd + 4 * hi;
+        lse4[qd] = load_f32x4_guard(lsep, qrow, S);
+        dl4[qd] = load_f32x4_guard(deltap, qrow, S);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float lse2 = lse4[r >> 2][r & 3] * LOG2E;
+        const float delt = dl4[r >> 2][r & 3];
+        float p;
+        if (edge) {
+          const int q_abs_r = q0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          p = (q_abs_r >= kv_abs && q_abs_r < S && kv_abs < S)
+                  ? exp2f(sa[r] * sc2 - lse2) : 0.f;
+        } else {
+          p = exp2f(sa[r] * sc2 - lse2);
+        }
+        pt_val[t][r] = p;
+        dst_val[t][r] = p * (dpa[r] - delt) * scale;
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    if (qt + 1 < n_q_tiles) {
+      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
+      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
+      tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+      tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+      if (qt + 2 < n_q_tiles) {
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+      }
+    }
+
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      const bf16x8 pa = pack_p_frag(pt_val[s >> 1], s & 1);
+      const bf16x8 dsa = pack_p_frag(dst_val[s >> 1], s & 1);
+#pragma unroll
+      for (int ht = 0; ht < NT32; ++ht) {
+        dv_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pa, ldsT_frag(dotb, ht * 32 + (lane & 31), s * 16 + hi * 8), dv_acc[ht], 0, 0, 0);
+        dk_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            dsa, ldsT_frag(qtb, ht * 32 + (lane & 31), s * 16 + hi * 8), dk_acc[ht], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+  }
+
+  // epilogue: lane holds one hd column x 16 kv rows per 32-tile
+#pragma unroll
+  for (int ht = 0; ht < NT32; ++ht) {
+    const int hd_c = 32 * ht + (lane & 31);
+    if (hd_c >= hd) continue;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kv_row = kv_start + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      if (kv_row < S) {
+        dk[base + (long)kv_row * hd + hd_c] = __float2bfloat16(dk_acc[ht][r]);
+        dv[base + (long)kv_row * hd + hd_c] = __float2bfloat16(dv_acc[ht][r]);
+      }
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -747,6 +1290,15 @@ static int pad32(int hd) { return (hd + 31) / 32 * 32; }
     }                                                             \
   } while (0)
 
+// RELORA_AMD_ATTN_FWD=2 falls back to the v2 (16x16 C-layout) forward
+static int attn_fwd_version() {
+  static int v = [] {
+    const char* e = getenv("RELORA_AMD_ATTN_FWD");
+    return (e && e[0] == '2') ? 2 : 3;
+  }();
+  return v;
+}
+
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     double scale) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous());
@@ -757,6 +1309,19 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto lse = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 block(512);
+  if (attn_fwd_version() == 3) {
+    DISPATCH_HD(HDP, {
+      dim3 grid((S + 255) / 256, B * nh);
+      const int LDK = HD + LPAD;
+      size_t smem = (2 * TILE * LDK + 2 * HD * TILE) * sizeof(__bf16);
+      hipLaunchKernelGGL((attn_fwd_v3_kernel<HD>), grid, block, smem, stream,
+                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
+                         lse.data_ptr<float>(), S, hd, (float)scale);
+    });
+    HIP_CHECK_LAST();
+    return {out, lse};
+  }
   DISPATCH_HD(HDP, {
     constexpr int RF = (HD <= 64) ? 1 : 2;
     dim3 grid((S + 128 * RF - 1) / (128 * RF), B * nh);
@@ -788,28 +1353,57 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                      delta.data_ptr<float>(), hd);
   HIP_CHECK_LAST();
 
+  // RELORA_AMD_ATTN_BWD=2 falls back to the v2 (16x16 C-layout) backward
+  static int bwd_ver = [] {
+    const char* e = getenv("RELORA_AMD_ATTN_BWD");
+    return (e && e[0] == '2') ? 2 : 3;
+  }();
+
   dim3 block(512);
   DISPATCH_HD(HDP, {
     const int LDK = HD + LPAD, LDT = TILE + LPAD;
-    size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
-    hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
-                       smem_dq, stream,
-                       (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
-    HIP_CHECK_LAST();
-    constexpr int NBUF = (HD <= 64) ? 2 : 1;
-    size_t smem_dkdv =
-        (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 2 * 8 * 16 * LDT) * sizeof(__bf16);
-    hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
-                       smem_dkdv, stream,
-                       (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
-                       (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
-                       S, hd, (float)scale);
-    HIP_CHECK_LAST();
+    if (bwd_ver == 3) {
+      size_t smem_dq3 = (2 * TILE * LDK * 2 + 2 * HD * TILE) * sizeof(__bf16);
+      hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
+                         smem_dq3, stream,
+                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),
+                         (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
+      HIP_CHECK_LAST();
+    } else {
+      size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
+      hipLaunchKernelGGL((attn_bwd_dq_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
+                         smem_dq, stream,
+                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),
+                         (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
+      HIP_CHECK_LAST();
+    }
+    if (bwd_ver == 3 && HD <= 64) {
+      size_t smem_dkdv3 = (2 * TILE * LDK * 2 + 2 * HD * TILE * 2) * sizeof(__bf16);
+      hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
+                         smem_dkdv3, stream,
+                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),
+                         (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
+                         S, hd, (float)scale);
+      HIP_CHECK_LAST();
+    } else {
+      constexpr int NBUF = (HD <= 64) ? 2 : 1;
+      size_t smem_dkdv =
+          (NBUF * TILE * LDK * 2 + NBUF * HD * TILE * 2 + 2 * 8 * 16 * LDT) * sizeof(__bf16);
+      hipLaunchKernelGGL((attn_bwd_dkdv_kernel<HD>), dim3((S + 127) / 128, B * nh), block,
+                         smem_dkdv, stream,
+                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                         lse.data_ptr<float>(), delta.data_ptr<float>(),
+                         (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
+                         S, hd, (float)scale);
+      HIP_CHECK_LAST();
+    }
   });
   return {dq, dk, dv};
 }
