@@ -446,8 +446,8 @@ DEV_INLINE bf16x8 pack_p_frag(const float* p16, int g) {
 #define LOG2E 1.44269504088896340736f
 #define DEFER_MAX_THR 11.5f  // log2 domain ~ e^8 (guide T13; bf16 accum headroom)
 
-template <int HD>
-__global__ __launch_bounds__(512) void attn_fwd_v3_kernel(
+template <int HD, int MINW = 2>  // MINW: min waves/SIMD hint (4 caps VGPRs at 128)
+__global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
     float* __restrict__ lse, int S, int hd, float scale) {
@@ -505,85 +505,90 @@ __global__ __launch_bounds__(512) void attn_fwd_v3_kernel(
     const int kv0 = kt * TILE;
     const __bf16* kb = lds_k + cur * TILE * LDK;
     const __bf16* vtb = lds_vt + cur * HD * TILE;
-    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
-    // S^T = K·Q^T over two 32-kv subtiles; lane ends with 32 score f32s
-    float p_val[2][16];
-    __builtin_amdgcn_s_setprio(1);
+    // two independent online-softmax updates per staged tile, one per 32-kv
+    // subtile: halves the live score registers (16 f32, not 32) — keeps the
+    // hd64 kernel under the 128-VGPR / 2-blocks-per-CU occupancy cliff
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
-      f32x16 acc = (f32x16)(0.f);
+      const int kv0s = kv0 + 32 * t;
+      // wave-uniform edge: the wave's lowest q row decides the masked path
+      const bool edge = (kv0s + 31 > q_start + wave * 32) || (kv0s + 32 > S);
+      if (kv0s > q_max_abs) break;  // fully-masked subtile (block-uniform)
+
+      float p_val[16];
+      __builtin_amdgcn_s_setprio(1);
+      {
+        f32x16 acc = (f32x16)(0.f);
 #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks)
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            lds_frag(kb, 32 * t + qcol, ks * 16 + hi * 8, LDK), qf[ks], acc, 0, 0, 0);
-      if (edge) {
+        for (int ks = 0; ks < KSTEPS; ++ks)
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              lds_frag(kb, 32 * t + qcol, ks * 16 + hi * 8, LDK), qf[ks], acc, 0, 0, 0);
+        if (edge) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          p_val[t][r] = (kv_abs > q_abs || kv_abs >= S) ? -INFINITY : acc[r] * sc2;
+          for (int r = 0; r < 16; ++r) {
+            const int kv_abs = kv0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            p_val[r] = (kv_abs > q_abs || kv_abs >= S) ? -INFINITY : acc[r] * sc2;
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) p_val[r] = acc[r] * sc2;
         }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) p_val[t][r] = acc[r] * sc2;
       }
-    }
-    __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_setprio(0);
 
-    // stage tile kt+1 while the VALU softmax runs (T14 split)
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+      // stage tile kt+1 between the two MFMA clusters (T14 split)
+      if (t == 0 && kt + 1 < n_tiles) {
+        tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+        tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+        if (kt + 2 < n_tiles) {
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+        }
       }
-    }
 
-    // online softmax, fully per-lane (q = lane&31)
-    float m_tile = p_val[0][0];
+      // online softmax, fully per-lane (q = lane&31)
+      float m_tile = p_val[0];
 #pragma unroll
-    for (int i = 1; i < 16; ++i) m_tile = fmaxf(m_tile, p_val[0][i]);
-#pragma unroll
-    for (int i = 0; i < 16; ++i) m_tile = fmaxf(m_tile, p_val[1][i]);
-    m_tile = swap_combine_max(m_tile);
+      for (int i = 1; i < 16; ++i) m_tile = fmaxf(m_tile, p_val[i]);
+      m_tile = swap_combine_max(m_tile);
 
-    // defer-max: only rescale when the tile max exceeds the running max by
-    // more than THR (wave-uniform branch; textbook-safe order — decision
-    // precedes this tile's exponentiation, l-update follows at same scale)
-    if (__any(m_tile > m2 + DEFER_MAX_THR) || m2 == -INFINITY) {
-      const float m_new = fmaxf(m2, m_tile);
-      const float alpha = (m_new == -INFINITY) ? 0.f : exp2f(m2 - m_new);
-      l_run *= alpha;
+      // defer-max: only rescale when the subtile max exceeds the running max
+      // by more than THR (textbook-safe order — decision precedes this
+      // subtile's exponentiation, l-update follows at the same scale)
+      if (__any(m_tile > m2 + DEFER_MAX_THR) || m2 == -INFINITY) {
+        const float m_new = fmaxf(m2, m_tile);
+        const float alpha = (m_new == -INFINITY) ? 0.f : exp2f(m2 - m_new);
+        l_run *= alpha;
 #pragma unroll
-      for (int t = 0; t < NT32; ++t)
+        for (int ht = 0; ht < NT32; ++ht)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[t][r] *= alpha;
-      m2 = m_new;
-    }
+          for (int r = 0; r < 16; ++r) o_acc[ht][r] *= alpha;
+        m2 = m_new;
+      }
 
-    float rsum = 0.f;
-#pragma unroll
-    for (int t = 0; t < 2; ++t)
+      float rsum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float p = (m2 == -INFINITY) ? 0.f : exp2f(p_val[t][r] - m2);
-        p_val[t][r] = p;
+        const float p = (m2 == -INFINITY) ? 0.f : exp2f(p_val[r] - m2);
+        p_val[r] = p;
         rsum += p;
       }
-    l_run += swap_combine_sum(rsum);
+      l_run += swap_combine_sum(rsum);
 
-    // PV swapped: O^T += V^T · P^T, 4 k-steps over the 64-kv tile
-    __builtin_amdgcn_s_setprio(1);
+      // PV swapped: O^T += V^T · P^T, 2 k-steps over the 32-kv subtile
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const bf16x8 pfrag = pack_p_frag(p_val[s >> 1], s & 1);
+      for (int g = 0; g < 2; ++g) {
+        const bf16x8 pfrag = pack_p_frag(p_val, g);
+        const int s = 2 * t + g;
 #pragma unroll
-      for (int ht = 0; ht < NT32; ++ht)
-        o_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            ldsT_frag(vtb, ht * 32 + qcol, s * 16 + hi * 8), pfrag, o_acc[ht], 0, 0, 0);
+        for (int ht = 0; ht < NT32; ++ht)
+          o_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ldsT_frag(vtb, ht * 32 + qcol, s * 16 + hi * 8), pfrag, o_acc[ht], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -613,16 +618,33 @@ __global__ __launch_bounds__(512) void attn_fwd_v3_kernel(
 // backward: delta preprocess
 // ---------------------------------------------------------------------------
 
-__global__ void attn_delta_kernel(const __hip_bfloat16* __restrict__ dout,
-                                  const __hip_bfloat16* __restrict__ o,
-                                  float* __restrict__ delta, int hd) {
-  __shared__ float scratch[16];
-  const long row = blockIdx.x;
+// delta = rowsum(dO * O): each wave handles 8 rows (hd <= 128), bf16x8
+// vector loads, intra-8-lane-group reduction — 256-thread blocks cover 32
+// rows each (the one-row-per-block version was 5% of backward time)
+__global__ __launch_bounds__(256) void attn_delta_kernel(
+    const __hip_bfloat16* __restrict__ dout, const __hip_bfloat16* __restrict__ o,
+    float* __restrict__ delta, long n_rows, int hd) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int seg = lane >> 3;       // 8 lanes per row
+  const int sub = lane & 7;
+  const long row = (long)blockIdx.x * 32 + wave * 8 + seg;
+  if (row >= n_rows) return;
   float acc = 0.f;
-  for (int i = threadIdx.x; i < hd; i += blockDim.x)
-    acc += to_f32(dout[row * hd + i]) * to_f32(o[row * hd + i]);
-  acc = block_reduce_sum(acc, scratch);
-  if (threadIdx.x == 0) delta[row] = acc;
+  for (int c = sub * 8; c < hd; c += 64) {
+    if (c + 8 <= hd) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(dout + row * hd + c);
+      const bf16x8 b = *reinterpret_cast<const bf16x8*>(o + row * hd + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += (float)a[j] * (float)b[j];
+    } else {
+      for (int j = c; j < hd; ++j)
+        acc += to_f32(dout[row * hd + j]) * to_f32(o[row * hd + j]);
+    }
+  }
+#pragma unroll
+  for (int m = 1; m < 8; m <<= 1) acc += __shfl_xor(acc, m);
+  if (sub == 0) delta[row] = acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -716,56 +738,63 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
     const __bf16* kb = lds_k + cur * TILE * LDK;
     const __bf16* vb = lds_v + cur * TILE * LDK;
     const __bf16* ktb = lds_kt + cur * HD * TILE;
-    const bool edge = (kv0 + TILE - 1 > q_start) || (kv0 + TILE > S);
 
-    float ds_val[2][16];
-    __builtin_amdgcn_s_setprio(1);
+    // per-32-kv subtile (halves live score registers, like the v3 forward)
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
-      f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
+      const int kv0s = kv0 + 32 * t;
+      const bool edge_t = (kv0s + 31 > q_start + wave * 32) || (kv0s + 32 > S);
+      if (kv0s > q_max_abs) break;
+
+      float ds_val[16];
+      __builtin_amdgcn_s_setprio(1);
+      {
+        f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
 #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) {
-        const int c0 = ks * 16 + hi * 8;
-        sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            lds_frag(kb, 32 * t + qcol, c0, LDK), qf[ks], sa, 0, 0, 0);
-        dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            lds_frag(vb, 32 * t + qcol, c0, LDK), dof[ks], dpa, 0, 0, 0);
-      }
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        float p;
-        if (edge) {
-          const int kv_abs = kv0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          p = (kv_abs <= q_abs && kv_abs < S && q_abs < S)
-                  ? exp2f(sa[r] * sc2 - lse2) : 0.f;
-        } else {
-          p = exp2f(sa[r] * sc2 - lse2);
+        for (int ks = 0; ks < KSTEPS; ++ks) {
+          const int c0 = ks * 16 + hi * 8;
+          sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              lds_frag(kb, 32 * t + qcol, c0, LDK), qf[ks], sa, 0, 0, 0);
+          dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              lds_frag(vb, 32 * t + qcol, c0, LDK), dof[ks], dpa, 0, 0, 0);
         }
-        ds_val[t][r] = p * (dpa[r] - delt) * scale;
-      }
-    }
-    __builtin_amdgcn_s_setprio(0);
-
-    if (kt + 1 < n_tiles) {
-      tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-      tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
-      tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
-      if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
-      }
-    }
-
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const bf16x8 dsf = pack_p_frag(ds_val[s >> 1], s & 1);
+        for (int r = 0; r < 16; ++r) {
+          float p;
+          if (edge_t) {
+            const int kv_abs = kv0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            p = (kv_abs <= q_abs && kv_abs < S && q_abs < S)
+                    ? exp2f(sa[r] * sc2 - lse2) : 0.f;
+          } else {
+            p = exp2f(sa[r] * sc2 - lse2);
+          }
+          ds_val[r] = p * (dpa[r] - delt) * scale;
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      if (t == 0 && kt + 1 < n_tiles) {
+        tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+        tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+        tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
+        if (kt + 2 < n_tiles) {
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+        }
+      }
+
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int ht = 0; ht < NT32; ++ht)
-        dq_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            ldsT_frag(ktb, ht * 32 + qcol, s * 16 + hi * 8), dsf, dq_acc[ht], 0, 0, 0);
+      for (int g = 0; g < 2; ++g) {
+        const bf16x8 dsf = pack_p_frag(ds_val, g);
+        const int s = 2 * t + g;
+#pragma unroll
+        for (int ht = 0; ht < NT32; ++ht)
+          dq_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              ldsT_frag(ktb, ht * 32 + qcol, s * 16 + hi * 8), dsf, dq_acc[ht], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -857,72 +886,78 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
     const __bf16* dob = lds_do + cur * TILE * LDK;
     const __bf16* qtb = lds_qt + cur * HD * TILE;
     const __bf16* dotb = lds_dot + cur * HD * TILE;
-    const bool edge = (q0 < kv_start + 255) || (q0 + TILE > S);
 
-    float pt_val[2][16], dst_val[2][16];
-    __builtin_amdgcn_s_setprio(1);
+    // per-32-q subtile: halves the live P/dS registers and loads lse/delta
+    // one quad at a time — this is what keeps the hd64 kernel from spilling
 #pragma unroll
     for (int t = 0; t < 2; ++t) {
-      f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
+      const int q0s = q0 + 32 * t;
+      const bool edge_t = (q0s < kv_start + 255) || (q0s + 32 > S);
+
+      float pt_val[16], dst_val[16];
+      __builtin_amdgcn_s_setprio(1);
+      {
+        f32x16 sa = (f32x16)(0.f), dpa = (f32x16)(0.f);
 #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) {
-        const int c0 = ks * 16 + hi * 8;
-        sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            lds_frag(qb, 32 * t + (lane & 31), c0, LDK), kf[ks], sa, 0, 0, 0);
-        dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            lds_frag(dob, 32 * t + (lane & 31), c0, LDK), vf[ks], dpa, 0, 0, 0);
-      }
-      // lse/delta for the C-layout q rows: quads are q-contiguous
-      f32x4 lse4[4], dl4[4];
-#pragma unroll
-      for (int qd = 0; qd < 4; ++qd) {
-        const long qrow = q0 + 32 * t + 8 * qd + 4 * hi;
-        lse4[qd] = load_f32x4_guard(lsep, qrow, S);
-        dl4[qd] = load_f32x4_guard(deltap, qrow, S);
-      }
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float lse2 = lse4[r >> 2][r & 3] * LOG2E;
-        const float delt = dl4[r >> 2][r & 3];
-        float p;
-        if (edge) {
-          const int q_abs_r = q0 + 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          p = (q_abs_r >= kv_abs && q_abs_r < S && kv_abs < S)
-                  ? exp2f(sa[r] * sc2 - lse2) : 0.f;
-        } else {
-          p = exp2f(sa[r] * sc2 - lse2);
+        for (int ks = 0; ks < KSTEPS; ++ks) {
+          const int c0 = ks * 16 + hi * 8;
+          sa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              lds_frag(qb, 32 * t + (lane & 31), c0, LDK), kf[ks], sa, 0, 0, 0);
+          dpa = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              lds_frag(dob, 32 * t + (lane & 31), c0, LDK), vf[ks], dpa, 0, 0, 0);
         }
-        pt_val[t][r] = p;
-        dst_val[t][r] = p * (dpa[r] - delt) * scale;
-      }
-    }
-    __builtin_amdgcn_s_setprio(0);
-
-    if (qt + 1 < n_q_tiles) {
-      tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
-      tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
-      tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
-      tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
-      if (qt + 2 < n_q_tiles) {
-        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
-      }
-    }
-
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {
-      const bf16x8 pa = pack_p_frag(pt_val[s >> 1], s & 1);
-      const bf16x8 dsa = pack_p_frag(dst_val[s >> 1], s & 1);
+        for (int qd = 0; qd < 4; ++qd) {
+          // lse/delta for this C-layout row quad (q-contiguous)
+          const long qrow = q0s + 8 * qd + 4 * hi;
+          const f32x4 lse4 = load_f32x4_guard(lsep, qrow, S);
+          const f32x4 dl4 = load_f32x4_guard(deltap, qrow, S);
 #pragma unroll
-      for (int ht = 0; ht < NT32; ++ht) {
-        dv_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            pa, ldsT_frag(dotb, ht * 32 + (lane & 31), s * 16 + hi * 8), dv_acc[ht], 0, 0, 0);
-        dk_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            dsa, ldsT_frag(qtb, ht * 32 + (lane & 31), s * 16 + hi * 8), dk_acc[ht], 0, 0, 0);
+          for (int i = 0; i < 4; ++i) {
+            const int r = 4 * qd + i;
+            const float lse2 = lse4[i] * LOG2E;
+            float p;
+            if (edge_t) {
+              const int q_abs_r = q0s + (r & 3) + 8 * (r >> 2) + 4 * hi;
+              p = (q_abs_r >= kv_abs && q_abs_r < S && kv_abs < S)
+                      ? exp2f(sa[r] * sc2 - lse2) : 0.f;
+            } else {
+              p = exp2f(sa[r] * sc2 - lse2);
+            }
+            pt_val[r] = p;
+            dst_val[r] = p * (dpa[r] - dl4[i]) * scale;
+          }
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
+
+      if (t == 0 && qt + 1 < n_q_tiles) {
+        tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
+        tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
+        tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+        tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+        if (qt + 2 < n_q_tiles) {
+          tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+        }
+      }
+
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        const bf16x8 pa = pack_p_frag(pt_val, g);
+        const bf16x8 dsa = pack_p_frag(dst_val, g);
+        const int s = 2 * t + g;
+#pragma unroll
+        for (int ht = 0; ht < NT32; ++ht) {
+          dv_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa, ldsT_frag(dotb, ht * 32 + (lane & 31), s * 16 + hi * 8), dv_acc[ht], 0, 0, 0);
+          dk_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dsa, ldsT_frag(qtb, ht * 32 + (lane & 31), s * 16 + hi * 8), dk_acc[ht], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -1314,10 +1349,21 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
       dim3 grid((S + 255) / 256, B * nh);
       const int LDK = HD + LPAD;
       size_t smem = (2 * TILE * LDK + 2 * HD * TILE) * sizeof(__bf16);
-      hipLaunchKernelGGL((attn_fwd_v3_kernel<HD>), grid, block, smem, stream,
-                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
-                         (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
-                         lse.data_ptr<float>(), S, hd, (float)scale);
+      static const bool occ4 = [] {
+        const char* e = getenv("RELORA_AMD_ATTN_OCC4");
+        return !(e && e[0] == '0');  // default ON (measured +15% at hd64)
+      }();
+      if (occ4 && HD <= 64) {
+        hipLaunchKernelGGL((attn_fwd_v3_kernel<HD, 4>), grid, block, smem, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
+                           lse.data_ptr<float>(), S, hd, (float)scale);
+      } else {
+        hipLaunchKernelGGL((attn_fwd_v3_kernel<HD>), grid, block, smem, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
+                           lse.data_ptr<float>(), S, hd, (float)scale);
+      }
     });
     HIP_CHECK_LAST();
     return {out, lse};
@@ -1348,9 +1394,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dout = dout.contiguous();
 
-  hipLaunchKernelGGL(attn_delta_kernel, dim3((long)B * nh * S), dim3(64), 0, stream,
+  const long n_rows = (long)B * nh * S;
+  hipLaunchKernelGGL(attn_delta_kernel, dim3((n_rows + 31) / 32), dim3(256), 0, stream,
                      (const __hip_bfloat16*)dout.data_ptr(), (const __hip_bfloat16*)o.data_ptr(),
-                     delta.data_ptr<float>(), hd);
+                     delta.data_ptr<float>(), n_rows, hd);
   HIP_CHECK_LAST();
 
   // RELORA_AMD_ATTN_BWD=2 falls back to the v2 (16x16 C-layout) backward
@@ -1362,7 +1409,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
   dim3 block(512);
   DISPATCH_HD(HDP, {
     const int LDK = HD + LPAD, LDT = TILE + LPAD;
-    if (bwd_ver == 3) {
+    if (bwd_ver == 3 && HD <= 96) {  // dq v3 at hd128 hits the VGPR cap
       size_t smem_dq3 = (2 * TILE * LDK * 2 + 2 * HD * TILE) * sizeof(__bf16);
       hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
                          smem_dq3, stream,
