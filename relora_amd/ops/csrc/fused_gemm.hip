@@ -1,0 +1,243 @@
+#include "hip/hip_runtime.h"
+// Fused frozen-GEMM + LoRA rank-r epilogue (K1+K2 — the BASELINE.json
+// north-star kernel):  y[M,N] = x[M,K]·W[N,K]^T (+bias) + s·(t[M,r]·Bw[N,r]^T)
+// where t = dropout(x)·A^T is produced by the existing skinny LoRA kernel.
+//
+// Replaces the composed  F.linear (hipBLASLt) + lora_add_nt_  pair on
+// aligned shapes: the composed form pays a full [M,N] bf16 read-modify-write
+// (the lora_add) on top of the library GEMM; here the rank-r update runs as
+// an MFMA epilogue on the resident accumulators (+r/K extra MFMA work, no
+// extra [M,N] traffic).
+//
+// Structure (CDNA4 guide §5 "glds vs register staging", verified tier):
+//   256x256 output tile, BK=64, 8 waves (2Mx4N), per-wave 128x64 output,
+//   mfma_f32_16x16x32_bf16; both operand tiles staged by
+//   global_load_lds_dwordx4 into double-buffered LDS with the st_16x32
+//   XOR swizzle applied on the per-lane GLOBAL source address (the LDS
+//   image is lane-linear, as glds requires); one vmcnt(0)+barrier per
+//   K-tile (the 2-buffer overlap pattern).  All LDS lives in ONE
+//   __shared__ array (a second __shared__ object de-pipelines glds).
+//
+// Constraints (host-checked): M%256==0, N%256==0, K%64==0, r%64==0 (or
+// r==0), bf16 contiguous row-major.  Unaligned shapes keep the composed
+// library path (llama_1b's intermediate 5461 stays composed; qkvo and the
+// whole of llama_250m/llama_7b qualify).
+
+#include <ATen/hip/HIPContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_g;
+typedef __attribute__((ext_vector_type(4))) float f32x4_g;
+
+// st_16x32 swizzle on a byte offset within a [rows][64] bf16 image
+// (128-B rows): byte bit5 ^= bit9 — spreads each ds_read_b128 lane group
+// over four 16-B slots (guide: bank-conflict 141x down, +35%).
+DEV_INLINE unsigned swz(unsigned o) { return o ^ (((o >> 9) & 1u) << 5); }
+
+// Stage one [256][64] bf16 tile (32 KB) into `image` via glds: 32 wave-level
+// 1-KiB pieces, 4 per wave.  The global source address carries the swizzle;
+// rows outside [0, rows_total) are redirected to row 0 (caller guarantees
+// alignment so this never happens on the standard path).
+DEV_INLINE void stage_tile_glds(__bf16* image, const __hip_bfloat16* gbase,
+                                long row0, long ld, int k0, int wave, int lane) {
+#pragma unroll
+  for (int p4 = 0; p4 < 4; ++p4) {
+    const int piece = wave * 4 + p4;
+    const unsigned d = piece * 1024u + lane * 16u;   // LDS dest byte
+    const unsigned s = swz(d);                       // source byte in tile
+    const int srow = s >> 7;                         // 128-B rows
+    const int scol_b = s & 127;
+    const __hip_bfloat16* src =
+        gbase + (row0 + srow) * ld + k0 + (scol_b >> 1);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(image) + (piece * 1024u) / 4,
+        16, 0, 0);
+  }
+}
+
+// read an A/B fragment (16x16x32 layout: lane -> row l&15, 8 k at (l>>4)*8)
+// from a swizzled [.][64] image; `row` relative to the image, k0 in [0,64)
+DEV_INLINE bf16x8_g frag_swz(const __bf16* image, int row, int k0) {
+  const unsigned o = (unsigned)row * 128u + (unsigned)k0 * 2u;
+  return *reinterpret_cast<const bf16x8_g*>((const char*)image + swz(o));
+}
+
+__global__ __launch_bounds__(512) void fused_lora_gemm_kernel(
+    const __hip_bfloat16* __restrict__ x,   // [M,K]
+    const __hip_bfloat16* __restrict__ w,   // [N,K]
+    const __hip_bfloat16* __restrict__ t,   // [M,r] or null
+    const __hip_bfloat16* __restrict__ bw,  // [N,r] or null
+    const __hip_bfloat16* __restrict__ bias,  // [N] or null
+    __hip_bfloat16* __restrict__ y,         // [M,N]
+    long M, long N, long K, int r, float lora_scale) {
+  // one __shared__ array only (glds pipeline rule)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = (__bf16*)smem;  // [2][A 256x64 | B 256x64] = 128 KiB
+
+  const int nbn = (int)(N >> 8);
+  const int bm = blockIdx.x / nbn;
+  const int bn = blockIdx.x % nbn;
+  const long m0 = (long)bm << 8;
+  const long n0 = (long)bn << 8;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 2;   // 0..1: wave m-position
+  const int wc = wave & 3;    // 0..3: wave n-position
+  const int fr = lane & 15;   // fragment row
+  const int fq = lane >> 4;   // fragment k-quarter / C row group
+
+  f32x4_g acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4_g{0.f, 0.f, 0.f, 0.f};
+
+  __bf16* bufA[2] = {lds, lds + 2 * 256 * 64};
+  __bf16* bufB[2] = {lds + 256 * 64, lds + 3 * 256 * 64};
+
+  const int KT = (int)(K >> 6);
+  stage_tile_glds(bufA[0], x, m0, K, 0, wave, lane);
+  stage_tile_glds(bufB[0], w, n0, K, 0, wave, lane);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) {
+      stage_tile_glds(bufA[cur ^ 1], x, m0, K, (kt + 1) << 6, wave, lane);
+      stage_tile_glds(bufB[cur ^ 1], w, n0, K, (kt + 1) << 6, wave, lane);
+    }
+    const __bf16* A = bufA[cur];
+    const __bf16* B = bufB[cur];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_g af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    // drain next tile's glds, then release the buffers we just read
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  // ---- LoRA rank-r epilogue: acc += s * t-tile @ Bw-tile^T ---------------
+  if (r > 0) {
+    const int RC = r >> 6;  // [256][64] chunks per operand
+    // r<=128: both operands fit the 4 buffer slots at once
+#pragma unroll 1
+    for (int c = 0; c < RC; ++c) {
+      stage_tile_glds(bufA[c & 1], t, m0, r, c << 6, wave, lane);
+      stage_tile_glds(bufB[c & 1], bw, n0, r, c << 6, wave, lane);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      const __bf16* A = bufA[c & 1];
+      const __bf16* B = bufB[c & 1];
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8_g af[8], bf[4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+      __syncthreads();
+    }
+    // fold the LoRA scale: epilogue computed acc_main + t·Bw^T; we need
+    // acc_main + s·t·Bw^T.  Instead of scaling inside (would scale the
+    // main GEMM too), the host pre-scales t by s — nothing to do here.
+  }
+
+  // ---- epilogue: stage each wave's 128x64 tile through its LDS share ----
+  // (per-element global stores are store-issue-bound; LDS re-read gives
+  // b128 row stores — DESIGN.md rule 6)
+  __syncthreads();
+  __bf16* mine = lds + wave * (128 * 64);  // 16 KiB per wave
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = i * 16 + fq * 4 + reg;  // within wave tile
+        const int col = j * 16 + fr;
+        float v = acc[i][j][reg];
+        if (bias) v += (float)bias[n0 + wc * 64 + col];
+        mine[row * 64 + col] = (__bf16)v;  // writes 2-way (col pairs share a
+                                           // dword), reads conflict-free b128
+      }
+  // no barrier needed: each wave reads only its own region
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    const int row = rr * 8 + (lane >> 3);          // 0..127
+    const int cb = (lane & 7) * 8;                 // 8-col chunk
+    const bf16x8_g vv = *reinterpret_cast<const bf16x8_g*>(mine + row * 64 + cb);
+    *reinterpret_cast<bf16x8_g*>(
+        y + (m0 + wr * 128 + row) * N + n0 + wc * 64 + cb) = vv;
+  }
+}
+
+torch::Tensor fused_lora_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor t,
+                              torch::Tensor bw, torch::Tensor bias, double lora_scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && w.scalar_type() == torch::kBFloat16);
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "fused_lora_gemm requires M%256==0, N%256==0, K%64==0");
+  int r = 0;
+  const bool has_lora = t.defined() && t.numel() > 0;
+  torch::Tensor t_scaled;
+  if (has_lora) {
+    TORCH_CHECK(bw.defined() && bw.is_contiguous() && t.is_contiguous());
+    r = (int)t.size(1);
+    TORCH_CHECK(t.size(0) == M && bw.size(0) == N && bw.size(1) == r);
+    TORCH_CHECK(r % 64 == 0 && r <= 256, "r must be a multiple of 64, <= 256");
+    // fold the LoRA scale into t (bf16 rounding of s*t matches the
+    // composed path, which also rounds t·B^T·s contributions in bf16)
+    t_scaled = (lora_scale == 1.0) ? t : (t * lora_scale).contiguous();
+  }
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int nbn = (int)(N >> 8);
+  dim3 grid((M >> 8) * nbn);
+  dim3 block(512);
+  size_t smem = 4 * 256 * 64 * sizeof(__bf16);  // 128 KiB
+  hipLaunchKernelGGL(fused_lora_gemm_kernel, grid, block, smem, stream,
+                     (const __hip_bfloat16*)x.data_ptr(),
+                     (const __hip_bfloat16*)w.data_ptr(),
+                     has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
+                     has_lora ? (const __hip_bfloat16*)bw.data_ptr() : nullptr,
+                     (bias.defined() && bias.numel())
+                         ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
+                     (__hip_bfloat16*)y.data_ptr(), M, N, K,
+                     has_lora ? r : 0, (float)lora_scale);
+  HIP_CHECK_LAST();
+  return y;
+}

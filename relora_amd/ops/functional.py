@@ -371,6 +371,21 @@ def _next_dropout_seed():
     return (torch.initial_seed() & 0x7FFFFFFFFFFF) ^ (_lora_seed_counter[0] * 0x9E3779B97F4A7C15)
 
 
+# RELORA_AMD_FUSED_K1=1 routes aligned shapes through the single-kernel
+# fused GEMM+LoRA (ops/csrc/fused_gemm.hip) instead of hipBLASLt+lora_add
+_FUSED_K1 = os.environ.get("RELORA_AMD_FUSED_K1", "0") == "1"
+
+
+def _use_fused_k1(x2d, weight, lora_A):
+    if not _FUSED_K1:
+        return False
+    M, K = x2d.shape
+    N = weight.shape[0]
+    r = lora_A.shape[0]
+    return (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+            and r % 64 == 0 and r <= 256)
+
+
 class _FusedLoRALinear(torch.autograd.Function):
     """GPU path: the rank-r update accumulates into the main GEMM's output
     via the MFMA lora_add kernels; dropout runs fused with a persisted
@@ -393,8 +408,15 @@ class _FusedLoRALinear(torch.autograd.Function):
         del xd  # dA re-applies the mask inline in skinny_grad; no need to
                 # persist the dropped-out copy (64 MB per flagship linear)
         bs = lora_B * scale                         # [N, r]
-        y = F.linear(x2d, weight, bias)
-        hip.ext().lora_add_nt_(y, t_u, bs)          # y += t_u @ bs^T
+        if _use_fused_k1(x2d, weight, lora_A):
+            # single-kernel K1+K2: main MFMA GEMM with the rank-r update as
+            # an epilogue on the resident accumulators (no [M,N] RMW pass)
+            y = hip.ext().fused_lora_gemm(
+                x2d, weight, t_u, lora_B,
+                bias if bias is not None else x2d.new_empty(0), scale)
+        else:
+            y = F.linear(x2d, weight, bias)
+            hip.ext().lora_add_nt_(y, t_u, bs)      # y += t_u @ bs^T
         ctx.save_for_backward(x2d, mask if mask is not None else x2d.new_empty(0),
                               t_u, weight, lora_A, bs)
         ctx.scale = scale

@@ -37,6 +37,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "adamw.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "lora_gemm.hip"),
+        os.path.join(CSRC, "fused_gemm.hip"),
         os.path.join(CSRC, "quantize.hip"),
     ],
     extra_compile_args={
