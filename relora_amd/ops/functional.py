@@ -371,19 +371,24 @@ def _next_dropout_seed():
     return (torch.initial_seed() & 0x7FFFFFFFFFFF) ^ (_lora_seed_counter[0] * 0x9E3779B97F4A7C15)
 
 
-# RELORA_AMD_FUSED_K1=1 routes aligned shapes through the single-kernel
-# fused GEMM+LoRA (ops/csrc/fused_gemm.hip) instead of hipBLASLt+lora_add
-_FUSED_K1 = os.environ.get("RELORA_AMD_FUSED_K1", "0") == "1"
+# The single-kernel fused GEMM+LoRA (ops/csrc/fused_gemm.hip) replaces
+# hipBLASLt+lora_add where the measured A/B favors it: aligned shapes with
+# K <= 1024 (profiles/fused_gemm_ab.log — fused wins +8..19% on the
+# llama_250m shapes; at K >= 2048 the library's deeper-pipelined GEMM wins
+# by 6-36%, so those stay composed).  RELORA_AMD_FUSED_K1=0 disables,
+# =all forces it on every aligned shape (for A/B runs).
+_FUSED_K1 = os.environ.get("RELORA_AMD_FUSED_K1", "1")
 
 
 def _use_fused_k1(x2d, weight, lora_A):
-    if not _FUSED_K1:
+    if _FUSED_K1 == "0":
         return False
     M, K = x2d.shape
     N = weight.shape[0]
     r = lora_A.shape[0]
-    return (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
-            and r % 64 == 0 and r <= 256)
+    aligned = (M % 256 == 0 and N % 256 == 0 and K % 64 == 0
+               and r % 64 == 0 and r <= 256)
+    return aligned and (K <= 1024 or _FUSED_K1 == "all")
 
 
 class _FusedLoRALinear(torch.autograd.Function):
