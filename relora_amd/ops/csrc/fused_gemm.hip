@@ -202,6 +202,209 @@ __global__ __launch_bounds__(512) void fused_lora_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// NF4 variant (K15): the frozen W arrives as packed NF4 (64-element blocks,
+// fp32 absmax, hi-nibble = even element — ops/csrc/quantize.hip layout) and
+// is dequantized DURING LDS staging, so no dense [N,K] W ever exists in HBM
+// (the reference runs bnb matmul_4bit, relora.py:314-317; the round-1 path
+// materialized the full dense W per forward — this kernel removes that).
+// Same tile structure; B staged by registers (dequant needs a transform,
+// which glds cannot do), A stays glds.
+// ---------------------------------------------------------------------------
+
+__constant__ float NF4_CODE_G[16] = {
+    -1.0f, -0.6961928009986877f, -0.5250730514526367f, -0.39491748809814453f,
+    -0.28444138169288635f, -0.18477343022823334f, -0.09105003625154495f, 0.0f,
+    0.07958029955625534f, 0.16093020141124725f, 0.24611230194568634f,
+    0.33791524171829224f, 0.44070982933044434f, 0.5626170039176941f,
+    0.7229568362236023f, 1.0f};
+
+// stage one [256][64] bf16 B-image from NF4-packed W: thread tid covers
+// row = tid/2, elements (tid&1)*32..+32 — one 16-byte packed load + one
+// absmax, 32 dequants, 4 swizzled ds_write_b128
+DEV_INLINE void stage_tile_nf4(__bf16* image, const uint8_t* qdata,
+                               const float* absmax, long row0, long K, int k0,
+                               int tid) {
+  const int row = tid >> 1;
+  const int half = tid & 1;
+  const long blk = (row0 + row) * (K >> 6) + (k0 >> 6);
+  const float am = absmax[blk];
+  const uint8_t* src = qdata + blk * 32 + half * 16;
+  union { uint32_t u[4]; uint8_t b[16]; } pk;
+  *reinterpret_cast<uint4*>(pk.u) = *reinterpret_cast<const uint4*>(src);
+  __bf16 vals[32];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    const uint8_t b = pk.b[i];
+    vals[2 * i] = (__bf16)(NF4_CODE_G[b >> 4] * am);
+    vals[2 * i + 1] = (__bf16)(NF4_CODE_G[b & 15] * am);
+  }
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const unsigned o = (unsigned)row * 128u + (unsigned)half * 64u + c * 16u;
+    *reinterpret_cast<bf16x8_g*>((char*)image + swz(o)) =
+        *reinterpret_cast<const bf16x8_g*>(vals + c * 8);
+  }
+}
+
+__global__ __launch_bounds__(512) void fused_nf4_gemm_kernel(
+    const __hip_bfloat16* __restrict__ x,   // [M,K]
+    const uint8_t* __restrict__ qw,         // packed NF4 of W [N,K]
+    const float* __restrict__ amax,         // absmax per 64-block
+    const __hip_bfloat16* __restrict__ t,   // [M,r] or null (pre-scaled)
+    const __hip_bfloat16* __restrict__ bw,  // [N,r] or null
+    const __hip_bfloat16* __restrict__ bias,
+    __hip_bfloat16* __restrict__ y, long M, long N, long K, int r) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = (__bf16*)smem;
+
+  const int nbn = (int)(N >> 8);
+  const int bm = blockIdx.x / nbn;
+  const int bn = blockIdx.x % nbn;
+  const long m0 = (long)bm << 8;
+  const long n0 = (long)bn << 8;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4_g acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4_g{0.f, 0.f, 0.f, 0.f};
+
+  __bf16* bufA[2] = {lds, lds + 2 * 256 * 64};
+  __bf16* bufB[2] = {lds + 256 * 64, lds + 3 * 256 * 64};
+
+  const int KT = (int)(K >> 6);
+  stage_tile_glds(bufA[0], x, m0, K, 0, wave, lane);
+  stage_tile_nf4(bufB[0], qw, amax, n0, K, 0, tid);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) {
+      stage_tile_glds(bufA[cur ^ 1], x, m0, K, (kt + 1) << 6, wave, lane);
+      stage_tile_nf4(bufB[cur ^ 1], qw, amax, n0, K, (kt + 1) << 6, tid);
+    }
+    const __bf16* A = bufA[cur];
+    const __bf16* B = bufB[cur];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_g af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+
+  if (r > 0) {
+    const int RC = r >> 6;
+#pragma unroll 1
+    for (int c = 0; c < RC; ++c) {
+      stage_tile_glds(bufA[c & 1], t, m0, r, c << 6, wave, lane);
+      stage_tile_glds(bufB[c & 1], bw, n0, r, c << 6, wave, lane);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      const __bf16* A = bufA[c & 1];
+      const __bf16* B = bufB[c & 1];
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8_g af[8], bf[4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+      __syncthreads();
+    }
+  }
+
+  __syncthreads();
+  __bf16* mine = lds + wave * (128 * 64);
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = i * 16 + fq * 4 + reg;
+        const int col = j * 16 + fr;
+        float v = acc[i][j][reg];
+        if (bias) v += (float)bias[n0 + wc * 64 + col];
+        mine[row * 64 + col] = (__bf16)v;
+      }
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    const int row = rr * 8 + (lane >> 3);
+    const int cb = (lane & 7) * 8;
+    const bf16x8_g vv = *reinterpret_cast<const bf16x8_g*>(mine + row * 64 + cb);
+    *reinterpret_cast<bf16x8_g*>(
+        y + (m0 + wr * 128 + row) * N + n0 + wc * 64 + cb) = vv;
+  }
+}
+
+torch::Tensor fused_nf4_gemm(torch::Tensor x, torch::Tensor qw, torch::Tensor amax,
+                             long N, torch::Tensor t, torch::Tensor bw,
+                             torch::Tensor bias, double lora_scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(qw.scalar_type() == torch::kUInt8 && amax.scalar_type() == torch::kFloat32);
+  const long M = x.size(0), K = x.size(1);
+  TORCH_CHECK(qw.numel() == N * K / 2, "packed NF4 size mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "fused_nf4_gemm requires M%256==0, N%256==0, K%64==0");
+  int r = 0;
+  const bool has_lora = t.defined() && t.numel() > 0;
+  torch::Tensor t_scaled;
+  if (has_lora) {
+    r = (int)t.size(1);
+    TORCH_CHECK(r % 64 == 0 && r <= 256);
+    t_scaled = (lora_scale == 1.0) ? t : (t * lora_scale).contiguous();
+  }
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid((M >> 8) * (N >> 8));
+  dim3 block(512);
+  size_t smem = 4 * 256 * 64 * sizeof(__bf16);
+  hipLaunchKernelGGL(fused_nf4_gemm_kernel, grid, block, smem, stream,
+                     (const __hip_bfloat16*)x.data_ptr(), qw.data_ptr<uint8_t>(),
+                     amax.data_ptr<float>(),
+                     has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
+                     has_lora ? (const __hip_bfloat16*)bw.data_ptr() : nullptr,
+                     (bias.defined() && bias.numel())
+                         ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
+                     (__hip_bfloat16*)y.data_ptr(), M, N, K, has_lora ? r : 0);
+  HIP_CHECK_LAST();
+  return y;
+}
+
 torch::Tensor fused_lora_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor t,
                               torch::Tensor bw, torch::Tensor bias, double lora_scale) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());

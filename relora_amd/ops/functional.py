@@ -457,6 +457,77 @@ class _FusedLoRALinear(torch.autograd.Function):
                 dA, dB, None, None, None)
 
 
+class _QuantizedLoRALinear(torch.autograd.Function):
+    """K15 path: frozen W stays NF4/int8-packed end to end.  Forward runs
+    the dequant-fused MFMA GEMM (aligned NF4 shapes) or a transient
+    materialize; backward re-dequantizes transiently for dX.  Unlike the
+    dense path, NO dense [N,K] W is ever saved for backward — the HBM
+    footprint of the frozen weights stays at the packed size (the entire
+    point of --quantize; reference runs bnb matmul_4bit, relora.py:314-317).
+    """
+
+    @staticmethod
+    def forward(ctx, x, bias, lora_A, lora_B, scale, dropout_p, training, qw):
+        in_shape = x.shape
+        K = in_shape[-1]
+        N = qw.out_features
+        x2d = x.contiguous().view(-1, K)
+        M = x2d.shape[0]
+        use_dropout = dropout_p > 0 and training
+        if use_dropout:
+            seed = _next_dropout_seed() & 0x7FFFFFFFFFFFFFFF
+            xd, mask = hip.ext().dropout_mask_fwd(x2d, dropout_p, seed)
+        else:
+            xd, mask = x2d, None
+        t_u = xd @ lora_A.t()
+        del xd
+        r = lora_A.shape[0]
+        empty = x2d.new_empty(0)
+        fused_ok = (qw.mode == "4bit" and M % 256 == 0 and N % 256 == 0
+                    and K % 64 == 0 and r % 64 == 0 and r <= 256
+                    and _FUSED_K1 != "0")
+        if fused_ok:
+            y = hip.ext().fused_nf4_gemm(
+                x2d, qw.qdata, qw.absmax, N, t_u, lora_B,
+                bias if bias is not None else empty, scale)
+        else:
+            w = qw.materialize(torch.bfloat16)  # transient; freed below
+            y = F.linear(x2d, w, bias)
+            del w
+            hip.ext().lora_add_nt_(y, t_u, lora_B * scale)
+        ctx.save_for_backward(x2d, mask if mask is not None else empty,
+                              t_u, lora_A, lora_B)
+        ctx.qw = qw
+        ctx.scale = scale
+        ctx.dropout_p = dropout_p if use_dropout else 0.0
+        ctx.has_bias = bias is not None
+        return y.view(*in_shape[:-1], N)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, mask, t_u, lora_A, lora_B = ctx.saved_tensors
+        scale, p = ctx.scale, ctx.dropout_p
+        qw = ctx.qw
+        N = qw.out_features
+        dy2d = dy.contiguous().view(-1, N)
+        bs = lora_B * scale
+        u_s = dy2d @ bs
+        w = qw.materialize(torch.bfloat16)  # transient dense W for dX only
+        dx = dy2d @ w
+        del w
+        hip.ext().lora_add_nn_(dx, u_s, lora_A,
+                               mask if p > 0 else mask.new_empty(0, dtype=torch.uint8),
+                               1.0 / (1.0 - p) if p > 0 else 1.0)
+        empty_mask = mask.new_empty(0, dtype=torch.uint8)
+        dA = hip.ext().skinny_grad(u_s, x2d, mask if p > 0 else empty_mask,
+                                   1.0 / (1.0 - p) if p > 0 else 1.0,
+                                   1.0, False, lora_A.dtype)
+        dB = hip.ext().skinny_grad(t_u, dy2d, empty_mask, 1.0, scale, True, lora_B.dtype)
+        dbias = dy2d.sum(0) if ctx.has_bias else None
+        return (dx.view(dy.shape[:-1] + (qw.in_features,)), dbias,
+                dA, dB, None, None, None, None)
+
+
 def _fused_ok(x, weight, lora_A, scale, lora_only):
     r = lora_A.shape[0]
     # odd in/out dims (llama_1b intermediate 5461) are CORRECT through the
@@ -470,7 +541,7 @@ def _fused_ok(x, weight, lora_A, scale, lora_only):
 
 
 def lora_linear(x, weight, bias, lora_A, lora_B, scale, dropout_p=0.0,
-                training=False, lora_only=False):
+                training=False, lora_only=False, quantized_weight=None):
     """y = x W^T (+b) + s * dropout(x) A^T B^T.
 
     GPU bf16 path: _FusedLoRALinear (MFMA rank-r accumulate kernels).
@@ -479,6 +550,15 @@ def lora_linear(x, weight, bias, lora_A, lora_B, scale, dropout_p=0.0,
     already passed through tanh by the caller) — tensor scale uses the
     composed path so autograd reaches it.
     """
+    if quantized_weight is not None:
+        if (hip.use_hip(x, "lora") and not lora_only and not torch.is_tensor(scale)
+                and x.dtype == torch.bfloat16):
+            return _QuantizedLoRALinear.apply(x, bias, lora_A, lora_B,
+                                              float(scale), dropout_p, training,
+                                              quantized_weight)
+        # CPU / tensor-scale fallback: transient materialize + composed ops
+        weight = quantized_weight.materialize(x.dtype if x.is_floating_point()
+                                              else torch.float32)
     if _fused_ok(x, weight, lora_A, scale, lora_only):
         return _FusedLoRALinear.apply(x, weight, bias, lora_A, lora_B,
                                       float(scale), dropout_p, training)

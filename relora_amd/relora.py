@@ -295,9 +295,13 @@ class ReLoRaLinear(nn.Module):
             nn.init.zeros_(self.scaling)
 
     def forward(self, x: torch.Tensor):
+        # quantized: pass the packed weight through — the GPU path runs the
+        # dequant-fused GEMM and never holds a dense [out,in] W (K15); the
+        # CPU path materializes transiently inside lora_linear
+        quantized = self.weight if self.quantize is not None else None
         return ops.lora_linear(
             x,
-            self._dense_weight(),
+            self.weight if self.quantize is None else None,
             self.bias,
             self.lora_A.weight,
             self.lora_B.weight,
@@ -305,6 +309,7 @@ class ReLoRaLinear(nn.Module):
             dropout_p=self.lora_dropout_p,
             training=self.training,
             lora_only=self.lora_only,
+            quantized_weight=quantized,
         )
 
     def extra_repr(self):
