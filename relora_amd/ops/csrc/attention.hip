@@ -92,14 +92,15 @@ DEV_INLINE bf16x8 lds_frag(const __bf16* tile, int row, int k0, int ldst) {
 
 // load an A fragment directly from global [S, hd]: lane row `grow`,
 // 8 k-elements at k0; zero-pad outside.
-DEV_INLINE bf16x8 global_frag(const __hip_bfloat16* src, int grow, int S, int hd, int k0) {
+DEV_INLINE bf16x8 global_frag(const __hip_bfloat16* src, int grow, int S, int hd,
+                              int ld, int k0) {
   bf16x8 v;
   if (grow < S && k0 + 8 <= hd) {
-    v = *reinterpret_cast<const bf16x8*>(src + (long)grow * hd + k0);
+    v = *reinterpret_cast<const bf16x8*>(src + (long)grow * ld + k0);
   } else {
 #pragma unroll
     for (int j = 0; j < 8; ++j)
-      v[j] = (grow < S && k0 + j < hd) ? (__bf16)src[(long)grow * hd + k0 + j] : (__bf16)0.f;
+      v[j] = (grow < S && k0 + j < hd) ? (__bf16)src[(long)grow * ld + k0 + j] : (__bf16)0.f;
   }
   return v;
 }
@@ -111,7 +112,7 @@ DEV_INLINE bf16x8 global_frag(const __hip_bfloat16* src, int grow, int S, int hd
 
 template <int HD, int NV>
 DEV_INLINE void tile_load_regs(bf16x8 (&r)[NV], const __hip_bfloat16* src,
-                               int row0, int S, int hd) {
+                               int row0, int S, int hd, int ld) {
   constexpr int C8 = HD / 8;
 #pragma unroll
   for (int i = 0; i < NV; ++i) {
@@ -119,7 +120,7 @@ DEV_INLINE void tile_load_regs(bf16x8 (&r)[NV], const __hip_bfloat16* src,
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
     const int c = (slot % C8) * 8;
-    r[i] = global_frag(src, row0 + row, S, hd, c);
+    r[i] = global_frag(src, row0 + row, S, hd, ld, c);
   }
 }
 
@@ -198,7 +199,8 @@ template <int HD>  // padded head dim (multiple of 32), actual hd passed in
 __global__ __launch_bounds__(512) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
-    float* __restrict__ lse, int S, int hd, float scale) {
+    float* __restrict__ lse, int S, int hd, int nh, long bst, long hst, int ld,
+    float scale) {
   constexpr int KFRAGS = HD / 32;      // QK^T k-steps
   constexpr int NT_HD = HD / 16;       // PV hd tiles
   constexpr int LDK = HD + LPAD;
@@ -219,7 +221,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   // heavy blocks (high q_start: up to 8x the kv tiles of block 0) first,
   // so the causal work imbalance doesn't leave a long tail
   const int q_start = (gridDim.x - 1 - blockIdx.x) * (128 * RF);
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -237,7 +239,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
 #pragma unroll
     for (int kf = 0; kf < KFRAGS; ++kf)
       qfrag[rf][kf] = global_frag(qp, q_start + rf * 128 + wave * 16 + col,
-                                  S, hd, kf * 32 + kgrp * 8);
+                                  S, hd, ld, kf * 32 + kgrp * 8);
 
   float m_run[RF][4], l_run[RF][4];
   f32x4 o_acc[RF][NT_HD];
@@ -257,13 +259,13 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
   const int n_tiles = (q_max_abs / TILE) + 1;  // causal bound
 
   bf16x8 rk[NV], rv[NV];
-  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
-  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_t<HD, NV>(lds_vt, rv);
   if (n_tiles > 1) {
-    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
-    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -314,8 +316,8 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
         tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
         tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
         if (kt + 2 < n_tiles) {
-          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
         }
       }
 
@@ -376,7 +378,7 @@ __global__ __launch_bounds__(512) void attn_fwd_kernel(
       for (int t = 0; t < NT_HD; ++t) {
         const int c = t * 16 + col;
         if (c < hd)
-          out[base + (long)row_abs * hd + c] = __float2bfloat16(o_acc[rf][t][reg] * inv_l);
+          out[base + (long)row_abs * ld + c] = __float2bfloat16(o_acc[rf][t][reg] * inv_l);
       }
       if (col == 0)
         lse[(long)bh * S + row_abs] = m_run[rf][reg] + __logf(fmaxf(l_run[rf][reg], 1e-30f));
@@ -450,7 +452,8 @@ template <int HD, int MINW = 2>  // MINW: min waves/SIMD hint (4 caps VGPRs at 1
 __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
-    float* __restrict__ lse, int S, int hd, float scale) {
+    float* __restrict__ lse, int S, int hd, int nh, long bst, long hst, int ld,
+    float scale) {
   constexpr int KSTEPS = HD / 16;  // QK^T k-steps per 32-kv subtile
   constexpr int NT32 = HD / 32;    // O^T 32-row hd tiles
   constexpr int LDK = HD + LPAD;
@@ -462,7 +465,7 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
 
   const int bh = blockIdx.y;
   const int q_start = (gridDim.x - 1 - blockIdx.x) * 256;  // heavy blocks first
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -477,7 +480,7 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
   bf16x8 qf[KSTEPS];
 #pragma unroll
   for (int ks = 0; ks < KSTEPS; ++ks)
-    qf[ks] = global_frag(qp, q_abs, S, hd, ks * 16 + hi * 8);
+    qf[ks] = global_frag(qp, q_abs, S, hd, ld, ks * 16 + hi * 8);
 
   float m2 = -INFINITY;  // running max of scores * scale * log2e
   float l_run = 0.f;
@@ -490,13 +493,13 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
   const float sc2 = scale * LOG2E;
 
   bf16x8 rk[NV], rv[NV];
-  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
-  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_t<HD, NV>(lds_vt, rv);
   if (n_tiles > 1) {
-    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
-    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -542,8 +545,8 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
         tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
         tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
         if (kt + 2 < n_tiles) {
-          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
         }
       }
 
@@ -603,9 +606,9 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
         if (hd_c + 1 < hd) {
           const unsigned w =
               pack_bf16(o_acc[ht][2 * d] * inv_l, o_acc[ht][2 * d + 1] * inv_l);
-          *reinterpret_cast<unsigned*>(out + base + (long)q_abs * hd + hd_c) = w;
+          *reinterpret_cast<unsigned*>(out + base + (long)q_abs * ld + hd_c) = w;
         } else if (hd_c < hd) {
-          out[base + (long)q_abs * hd + hd_c] = __float2bfloat16(o_acc[ht][2 * d] * inv_l);
+          out[base + (long)q_abs * ld + hd_c] = __float2bfloat16(o_acc[ht][2 * d] * inv_l);
         }
       }
     if (hi == 0)
@@ -623,23 +626,28 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
 // rows each (the one-row-per-block version was 5% of backward time)
 __global__ __launch_bounds__(256) void attn_delta_kernel(
     const __hip_bfloat16* __restrict__ dout, const __hip_bfloat16* __restrict__ o,
-    float* __restrict__ delta, long n_rows, int hd) {
+    float* __restrict__ delta, long n_rows, int hd, int nh, int S,
+    long bst, long hst, int ld) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int seg = lane >> 3;       // 8 lanes per row
   const int sub = lane & 7;
-  const long row = (long)blockIdx.x * 32 + wave * 8 + seg;
+  const long row = (long)blockIdx.x * 32 + wave * 8 + seg;  // (b*nh + h)*S + s
   if (row >= n_rows) return;
+  const long s = row % S;
+  const long h = (row / S) % nh;
+  const long b = row / ((long)S * nh);
+  const long off = b * bst + h * hst + s * ld;
   float acc = 0.f;
   for (int c = sub * 8; c < hd; c += 64) {
     if (c + 8 <= hd) {
-      const bf16x8 a = *reinterpret_cast<const bf16x8*>(dout + row * hd + c);
-      const bf16x8 b = *reinterpret_cast<const bf16x8*>(o + row * hd + c);
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(dout + off + c);
+      const bf16x8 bb = *reinterpret_cast<const bf16x8*>(o + off + c);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc += (float)a[j] * (float)b[j];
+      for (int j = 0; j < 8; ++j) acc += (float)a[j] * (float)bb[j];
     } else {
       for (int j = c; j < hd; ++j)
-        acc += to_f32(dout[row * hd + j]) * to_f32(o[row * hd + j]);
+        acc += to_f32(dout[off + j]) * to_f32(o[off + j]);
     }
   }
 #pragma unroll
@@ -678,7 +686,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dq, int S, int hd, float scale) {
+    __hip_bfloat16* __restrict__ dq, int S, int hd, int nh, long bst, long hst,
+    int ld, float scale) {
   constexpr int KSTEPS = HD / 16;
   constexpr int NT32 = HD / 32;
   constexpr int LDK = HD + LPAD;
@@ -691,7 +700,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
 
   const int bh = blockIdx.y;
   const int q_start = (gridDim.x - 1 - blockIdx.x) * 256;  // heavy blocks first
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -706,8 +715,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
   bf16x8 qf[KSTEPS], dof[KSTEPS];
 #pragma unroll
   for (int ks = 0; ks < KSTEPS; ++ks) {
-    qf[ks] = global_frag(qp, q_abs, S, hd, ks * 16 + hi * 8);
-    dof[ks] = global_frag(dop, q_abs, S, hd, ks * 16 + hi * 8);
+    qf[ks] = global_frag(qp, q_abs, S, hd, ld, ks * 16 + hi * 8);
+    dof[ks] = global_frag(dop, q_abs, S, hd, ld, ks * 16 + hi * 8);
   }
   const float sc2 = scale * LOG2E;
   const float lse2 = (q_abs < S) ? lse[(long)bh * S + q_abs] * LOG2E : 0.f;
@@ -721,14 +730,14 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
   const int n_tiles = (q_max_abs / TILE) + 1;
 
   bf16x8 rk[NV], rv[NV];
-  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
-  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_rows<HD, NV>(lds_v, rv, LDK);
   tile_write_t<HD, NV>(lds_kt, rk);
   if (n_tiles > 1) {
-    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
-    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -778,8 +787,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
         tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
         tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
         if (kt + 2 < n_tiles) {
-          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
         }
       }
 
@@ -806,9 +815,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
         const int hd_c = ((2 * d) & 3) + 8 * ((2 * d) >> 2) + 4 * hi + 32 * ht;
         if (hd_c + 1 < hd) {
           const unsigned w = pack_bf16(dq_acc[ht][2 * d], dq_acc[ht][2 * d + 1]);
-          *reinterpret_cast<unsigned*>(dq + base + (long)q_abs * hd + hd_c) = w;
+          *reinterpret_cast<unsigned*>(dq + base + (long)q_abs * ld + hd_c) = w;
         } else if (hd_c < hd) {
-          dq[base + (long)q_abs * hd + hd_c] = __float2bfloat16(dq_acc[ht][2 * d]);
+          dq[base + (long)q_abs * ld + hd_c] = __float2bfloat16(dq_acc[ht][2 * d]);
         }
       }
   }
@@ -820,7 +829,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
-    int S, int hd, float scale) {
+    int S, int hd, int nh, long bst, long hst, int ld, float scale) {
   constexpr int KSTEPS = HD / 16;
   constexpr int NT32 = HD / 32;
   constexpr int LDK = HD + LPAD;
@@ -834,7 +843,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
 
   const int bh = blockIdx.y;
   const int kv_start = blockIdx.x * 256;
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -851,8 +860,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
   bf16x8 kf[KSTEPS], vf[KSTEPS];
 #pragma unroll
   for (int ks = 0; ks < KSTEPS; ++ks) {
-    kf[ks] = global_frag(kp, kv_abs, S, hd, ks * 16 + hi * 8);
-    vf[ks] = global_frag(vp, kv_abs, S, hd, ks * 16 + hi * 8);
+    kf[ks] = global_frag(kp, kv_abs, S, hd, ld, ks * 16 + hi * 8);
+    vf[ks] = global_frag(vp, kv_abs, S, hd, ld, ks * 16 + hi * 8);
   }
   const float sc2 = scale * LOG2E;
 
@@ -867,15 +876,15 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
   const int n_q_tiles = (S + TILE - 1) / TILE;
 
   bf16x8 rq[NV], rdo[NV];
-  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd);
-  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
+  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd, ld);
+  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd, ld);
   tile_write_rows<HD, NV>(lds_q, rq, LDK);
   tile_write_rows<HD, NV>(lds_do, rdo, LDK);
   tile_write_t<HD, NV>(lds_qt, rq);
   tile_write_t<HD, NV>(lds_dot, rdo);
   if (first_qt + 1 < n_q_tiles) {
-    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd);
-    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd);
+    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -937,8 +946,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
         tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
         tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
         if (qt + 2 < n_q_tiles) {
-          tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
-          tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+          tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd, ld);
         }
       }
 
@@ -970,8 +979,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
     for (int r = 0; r < 16; ++r) {
       const int kv_row = kv_start + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
       if (kv_row < S) {
-        dk[base + (long)kv_row * hd + hd_c] = __float2bfloat16(dk_acc[ht][r]);
-        dv[base + (long)kv_row * hd + hd_c] = __float2bfloat16(dv_acc[ht][r]);
+        dk[base + (long)kv_row * ld + hd_c] = __float2bfloat16(dk_acc[ht][r]);
+        dv[base + (long)kv_row * ld + hd_c] = __float2bfloat16(dv_acc[ht][r]);
       }
     }
   }
@@ -986,7 +995,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dq, int S, int hd, float scale) {
+    __hip_bfloat16* __restrict__ dq, int S, int hd, int nh, long bst, long hst,
+    int ld, float scale) {
   constexpr int KFRAGS = HD / 32;
   constexpr int NT_HD = HD / 16;
   constexpr int LDK = HD + LPAD;
@@ -1001,7 +1011,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 
   const int bh = blockIdx.y;
   const int q_start = (gridDim.x - 1 - blockIdx.x) * 128;  // heavy blocks first
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -1016,8 +1026,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   bf16x8 qfrag[KFRAGS], dofrag[KFRAGS];
 #pragma unroll
   for (int kf = 0; kf < KFRAGS; ++kf) {
-    qfrag[kf] = global_frag(qp, qrow_abs, S, hd, kf * 32 + kgrp * 8);
-    dofrag[kf] = global_frag(dop, qrow_abs, S, hd, kf * 32 + kgrp * 8);
+    qfrag[kf] = global_frag(qp, qrow_abs, S, hd, ld, kf * 32 + kgrp * 8);
+    dofrag[kf] = global_frag(dop, qrow_abs, S, hd, ld, kf * 32 + kgrp * 8);
   }
   // per-reg row stats
   float lse_r[4], delta_r[4];
@@ -1036,14 +1046,14 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int n_tiles = (q_max_abs / TILE) + 1;
 
   bf16x8 rk[NV], rv[NV];
-  tile_load_regs<HD, NV>(rk, kp, 0, S, hd);
-  tile_load_regs<HD, NV>(rv, vp, 0, S, hd);
+  tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
+  tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
   tile_write_rows<HD, NV>(lds_v, rv, LDK);
   tile_write_t<HD, NV>(lds_kt, rk);
   if (n_tiles > 1) {
-    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd);
-    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd);
+    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -1089,8 +1099,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
       tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
       tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
       if (kt + 2 < n_tiles) {
-        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
+        tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
       }
     }
 
@@ -1123,7 +1133,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     for (int t = 0; t < NT_HD; ++t) {
       const int c = t * 16 + col;
       if (c < hd)
-        dq[base + (long)row_abs * hd + c] = __float2bfloat16(dq_acc[t][reg]);
+        dq[base + (long)row_abs * ld + c] = __float2bfloat16(dq_acc[t][reg]);
     }
   }
 }
@@ -1138,7 +1148,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv,
-    int S, int hd, float scale) {
+    int S, int hd, int nh, long bst, long hst, int ld, float scale) {
   constexpr int KFRAGS = HD / 32;
   constexpr int NT_HD = HD / 16;
   constexpr int LDK = HD + LPAD;
@@ -1157,7 +1167,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
 
   const int bh = blockIdx.y;
   const int kv_start_blk = blockIdx.x * 128;
-  const long base = (long)bh * S * hd;
+  const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
   const __hip_bfloat16* vp = v + base;
@@ -1172,8 +1182,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   bf16x8 kfrag[KFRAGS], vfrag[KFRAGS];
 #pragma unroll
   for (int kf = 0; kf < KFRAGS; ++kf) {
-    kfrag[kf] = global_frag(kp, kvrow_abs, S, hd, kf * 32 + kgrp * 8);
-    vfrag[kf] = global_frag(vp, kvrow_abs, S, hd, kf * 32 + kgrp * 8);
+    kfrag[kf] = global_frag(kp, kvrow_abs, S, hd, ld, kf * 32 + kgrp * 8);
+    vfrag[kf] = global_frag(vp, kvrow_abs, S, hd, ld, kf * 32 + kgrp * 8);
   }
 
   f32x4 dk_acc[NT_HD], dv_acc[NT_HD];
@@ -1187,15 +1197,15 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
   const int n_q_tiles = (S + TILE - 1) / TILE;
 
   bf16x8 rq[NV], rdo[NV];
-  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd);
-  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd);
+  tile_load_regs<HD, NV>(rq, qp, first_qt * TILE, S, hd, ld);
+  tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd, ld);
   tile_write_rows<HD, NV>(lds_q, rq, LDK);
   tile_write_rows<HD, NV>(lds_do, rdo, LDK);
   tile_write_t<HD, NV>(lds_qt, rq);
   tile_write_t<HD, NV>(lds_dot, rdo);
   if (first_qt + 1 < n_q_tiles) {
-    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd);
-    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd);
+    tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd, ld);
+    tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -1247,8 +1257,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
       tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
       if (qt + 2 < n_q_tiles) {
-        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd, ld);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd, ld);
       }
     }
 
@@ -1286,8 +1296,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
       tile_write_t<HD, NV>(lds_qt, rq);
       tile_write_t<HD, NV>(lds_dot, rdo);
       if (qt + 2 < n_q_tiles) {
-        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd);
-        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd);
+        tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd, ld);
+        tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd, ld);
       }
       __syncthreads();
     }
@@ -1301,8 +1311,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_kernel(
     for (int t = 0; t < NT_HD; ++t) {
       const int c = t * 16 + col;
       if (c < hd) {
-        dk[base + (long)row_abs * hd + c] = __float2bfloat16(dk_acc[t][reg]);
-        dv[base + (long)row_abs * hd + c] = __float2bfloat16(dv_acc[t][reg]);
+        dk[base + (long)row_abs * ld + c] = __float2bfloat16(dk_acc[t][reg]);
+        dv[base + (long)row_abs * ld + c] = __float2bfloat16(dv_acc[t][reg]);
       }
     }
   }
@@ -1334,13 +1344,37 @@ static int attn_fwd_version() {
   return v;
 }
 
+
+// Accept [B,nh,S,hd] tensors whose memory is either BHSD-contiguous or a
+// transposed view of a BSHD-contiguous buffer (what
+// `x.view(B,S,nh,hd).transpose(1,2)` yields — lets the model skip every
+// permute+contiguous copy around attention).  Returns {bst, hst, ld}.
+static std::array<long, 3> attn_strides(const torch::Tensor& q) {
+  const long B = q.size(0), nh = q.size(1), S = q.size(2), hd = q.size(3);
+  TORCH_CHECK(q.stride(3) == 1, "attention: head_dim must be innermost");
+  const bool bhsd = q.is_contiguous();
+  const bool bshd = (q.stride(1) == hd && q.stride(2) == nh * hd &&
+                     q.stride(0) == S * nh * hd);
+  TORCH_CHECK(bhsd || bshd, "attention: unsupported layout");
+  return {q.stride(0), q.stride(1), q.stride(2)};
+}
+
+static bool same_strides(const torch::Tensor& a, const torch::Tensor& b) {
+  return a.strides() == b.strides();
+}
+
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                     double scale) {
-  TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.is_contiguous());
+  TORCH_CHECK(q.is_cuda() && q.dim() == 4);
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16, "attn kernels are bf16-only");
   const int B = q.size(0), nh = q.size(1), S = q.size(2), hd = q.size(3);
   const int HDP = pad32(hd);
-  auto out = torch::empty_like(q);
+  const auto st = attn_strides(q);
+  TORCH_CHECK(same_strides(q, k) && same_strides(q, v),
+              "q/k/v must share a layout");
+  const long bst = st[0], hst = st[1];
+  const int ld = (int)st[2];
+  auto out = torch::empty_strided(q.sizes(), q.strides(), q.options());
   auto lse = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   dim3 block(512);
@@ -1357,12 +1391,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
         hipLaunchKernelGGL((attn_fwd_v3_kernel<HD, 4>), grid, block, smem, stream,
                            (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                            (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
-                           lse.data_ptr<float>(), S, hd, (float)scale);
+                           lse.data_ptr<float>(), S, hd, nh, bst, hst, ld, (float)scale);
       } else {
         hipLaunchKernelGGL((attn_fwd_v3_kernel<HD>), grid, block, smem, stream,
                            (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                            (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
-                           lse.data_ptr<float>(), S, hd, (float)scale);
+                           lse.data_ptr<float>(), S, hd, nh, bst, hst, ld, (float)scale);
       }
     });
     HIP_CHECK_LAST();
@@ -1376,7 +1410,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
     hipLaunchKernelGGL((attn_fwd_kernel<HD>), grid, block, smem, stream,
                        (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                        (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
-                       lse.data_ptr<float>(), S, hd, (float)scale);
+                       lse.data_ptr<float>(), S, hd, nh, bst, hst, ld, (float)scale);
   });
   HIP_CHECK_LAST();
   return {out, lse};
@@ -1387,17 +1421,27 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                                     double scale) {
   const int B = q.size(0), nh = q.size(1), S = q.size(2), hd = q.size(3);
   const int HDP = pad32(hd);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  const auto st = attn_strides(q);
+  const long bst = st[0], hst = st[1];
+  const int ld = (int)st[2];
+  // every tensor the kernels touch must share q's layout; copy dout into
+  // it if the incoming grad doesn't (o always does — attn_fwd allocated it)
+  TORCH_CHECK(same_strides(q, k) && same_strides(q, v) && same_strides(q, o));
+  if (!(dout.stride(3) == 1 && same_strides(q, dout))) {
+    auto d2 = torch::empty_strided(q.sizes(), q.strides(), q.options());
+    d2.copy_(dout);
+    dout = d2;
+  }
+  auto dq = torch::empty_strided(q.sizes(), q.strides(), q.options());
+  auto dk = torch::empty_strided(q.sizes(), q.strides(), q.options());
+  auto dv = torch::empty_strided(q.sizes(), q.strides(), q.options());
   auto delta = torch::empty({B, nh, S}, q.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  dout = dout.contiguous();
 
   const long n_rows = (long)B * nh * S;
   hipLaunchKernelGGL(attn_delta_kernel, dim3((n_rows + 31) / 32), dim3(256), 0, stream,
                      (const __hip_bfloat16*)dout.data_ptr(), (const __hip_bfloat16*)o.data_ptr(),
-                     delta.data_ptr<float>(), n_rows, hd);
+                     delta.data_ptr<float>(), n_rows, hd, nh, S, bst, hst, ld);
   HIP_CHECK_LAST();
 
   // RELORA_AMD_ATTN_BWD=2 falls back to the v2 (16x16 C-layout) backward
@@ -1416,7 +1460,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                          (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                          (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                          lse.data_ptr<float>(), delta.data_ptr<float>(),
-                         (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
+                         (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
+                         (float)scale);
       HIP_CHECK_LAST();
     } else {
       size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
@@ -1425,7 +1470,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                          (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                          (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                          lse.data_ptr<float>(), delta.data_ptr<float>(),
-                         (__hip_bfloat16*)dq.data_ptr(), S, hd, (float)scale);
+                         (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
+                         (float)scale);
       HIP_CHECK_LAST();
     }
     if (bwd_ver == 3 && HD <= 64) {
@@ -1436,7 +1482,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                          (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                          lse.data_ptr<float>(), delta.data_ptr<float>(),
                          (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
-                         S, hd, (float)scale);
+                         S, hd, nh, bst, hst, ld, (float)scale);
       HIP_CHECK_LAST();
     } else {
       constexpr int NBUF = (HD <= 64) ? 2 : 1;
@@ -1448,7 +1494,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
                          (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                          lse.data_ptr<float>(), delta.data_ptr<float>(),
                          (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
-                         S, hd, (float)scale);
+                         S, hd, nh, bst, hst, ld, (float)scale);
       HIP_CHECK_LAST();
     }
   });

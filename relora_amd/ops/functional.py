@@ -140,8 +140,10 @@ def rope_torch(q, k, cos, sin, position_ids=None):
 class _HipRoPE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, cos, sin):
-        q = q.contiguous()
-        k = k.contiguous()
+        # the kernel reads BHSD-contiguous AND transposed-BSHD views directly
+        # (same stride support as attention — no copies on the q/k path)
+        if not (_attn_layout_ok(q) and q.stride() == k.stride()):
+            q, k = q.contiguous(), k.contiguous()
         # Kernel contract: fp32 contiguous tables (a model .to(bf16) casts
         # registered buffers; converting [S,hd] back is noise next to the GEMMs).
         cos = cos.float().contiguous()
@@ -153,7 +155,9 @@ class _HipRoPE(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dqo, dko):
         cos, sin = ctx.saved_tensors
-        dq, dk = hip.ext().rope_fwd(dqo.contiguous(), dko.contiguous(), cos, sin, True)
+        if not (_attn_layout_ok(dqo) and dqo.stride() == dko.stride()):
+            dqo, dko = dqo.contiguous(), dko.contiguous()
+        dq, dk = hip.ext().rope_fwd(dqo, dko, cos, sin, True)
         return dq, dk, None, None
 
 
@@ -206,12 +210,23 @@ def sdpa_torch(q, k, v, causal=True, dropout_p=0.0):
     return F.scaled_dot_product_attention(q, k, v, dropout_p=dropout_p, is_causal=causal)
 
 
+def _attn_layout_ok(t):
+    """BHSD-contiguous, or a transposed view of a BSHD buffer (what the
+    model's `.view(B,S,nh,hd).transpose(1,2)` produces) — the kernels read
+    both directly, so no permute+contiguous copies happen around attention."""
+    if t.stride(-1) != 1:
+        return False
+    B, nh, S, hd = t.shape
+    s = tuple(t.stride())
+    return (t.is_contiguous()
+            or s == (S * nh * hd, hd, nh * hd, 1))
+
+
 class _HipFlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
-        q = q.contiguous()
-        k = k.contiguous()
-        v = v.contiguous()
+        if not (_attn_layout_ok(q) and q.stride() == k.stride() == v.stride()):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
         o, lse = hip.ext().attn_fwd(q, k, v, scale)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
@@ -220,7 +235,8 @@ class _HipFlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = hip.ext().attn_bwd(q, k, v, o, lse, do.contiguous(), ctx.scale)
+        # the host copies `do` into q's layout only if it differs
+        dq, dk, dv = hip.ext().attn_bwd(q, k, v, o, lse, do, ctx.scale)
         return dq, dk, dv, None
 
 
@@ -529,14 +545,19 @@ class _QuantizedLoRALinear(torch.autograd.Function):
 
 
 def _fused_ok(x, weight, lora_A, scale, lora_only):
+    if weight is None:
+        return False
     r = lora_A.shape[0]
     # odd in/out dims (llama_1b intermediate 5461) are CORRECT through the
-    # fused kernels (alignment-guarded fallbacks) but measured slower than
-    # the hipBLASLt composition — keep them on the library path
+    # fused kernels (alignment-guarded fallbacks); RELORA_AMD_LORA_ODD
+    # selects their path: "fused" (default — the composed torch path costs
+    # ~5 extra [M,N] passes per call: dropout, mul, add; see
+    # profiles/stage_r2 at::native attribution) or "torch"
+    dims_ok = (x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+               or os.environ.get("RELORA_AMD_LORA_ODD", "fused") == "fused")
     return (hip.use_hip(x, "lora") and not lora_only and not torch.is_tensor(scale)
             and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
-            and r % 32 == 0 and r <= 256
-            and x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
+            and r % 32 == 0 and r <= 256 and dims_ok
             and os.environ.get("RELORA_AMD_LORA_PATH", "fused") != "torch")
 
 
@@ -563,6 +584,15 @@ def lora_linear(x, weight, bias, lora_A, lora_B, scale, dropout_p=0.0,
         return _FusedLoRALinear.apply(x, weight, bias, lora_A, lora_B,
                                       float(scale), dropout_p, training)
     xd = F.dropout(x, p=dropout_p, training=training) if dropout_p > 0 else x
+    if x.is_cuda:
+        # fold the scale into the [M,r] intermediate (45x smaller than the
+        # [M,N] product the naive composition scales) — same math, removes
+        # a full-size elementwise pass; tensor scale keeps autograd
+        lora_out = F.linear(F.linear(xd, lora_A) * scale, lora_B)
+        if lora_only:
+            return lora_out
+        return F.linear(x, weight, bias).add_(lora_out)
+    # CPU path stays the bit-exact reference composition (parity oracle)
     lora_out = F.linear(F.linear(xd, lora_A), lora_B)
     if lora_only:
         return lora_out * scale
