@@ -549,12 +549,13 @@ def _fused_ok(x, weight, lora_A, scale, lora_only):
         return False
     r = lora_A.shape[0]
     # odd in/out dims (llama_1b intermediate 5461) are CORRECT through the
-    # fused kernels (alignment-guarded fallbacks); RELORA_AMD_LORA_ODD
-    # selects their path: "fused" (default — the composed torch path costs
-    # ~5 extra [M,N] passes per call: dropout, mul, add; see
-    # profiles/stage_r2 at::native attribution) or "torch"
+    # fused kernels (alignment-guarded fallbacks) but measured slower e2e
+    # than the composed path even after its scale-folding trim (58.7k vs
+    # 66.0k tok/s on the flagship, gpurun r2c A/B) — unaligned rows defeat
+    # every vectorized RMW path.  RELORA_AMD_LORA_ODD=fused re-routes them
+    # for future A/Bs.
     dims_ok = (x.shape[-1] % 8 == 0 and weight.shape[0] % 8 == 0
-               or os.environ.get("RELORA_AMD_LORA_ODD", "fused") == "fused")
+               or os.environ.get("RELORA_AMD_LORA_ODD", "torch") == "fused")
     return (hip.use_hip(x, "lora") and not lora_only and not torch.is_tensor(scale)
             and x.dtype == torch.bfloat16 and weight.dtype == torch.bfloat16
             and r % 32 == 0 and r <= 256 and dims_ok
