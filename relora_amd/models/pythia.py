@@ -143,7 +143,7 @@ class GPTNeoXMLP(nn.Module):
 
     def forward(self, hidden_states):
         hidden_states = self.dense_h_to_4h(hidden_states)
-        hidden_states = F.gelu(hidden_states)
+        hidden_states = ops.gelu(hidden_states)  # K8 HIP kernel on GPU
         return self.dense_4h_to_h(hidden_states)
 
 

@@ -17,6 +17,8 @@ std::vector<torch::Tensor> rope_fwd(torch::Tensor q, torch::Tensor k,
                                     bool inverse);
 // swiglu.hip
 torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u);
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor x, torch::Tensor dy);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u, torch::Tensor dy);
 // ce.hip
 std::vector<torch::Tensor> ce_row_stats(torch::Tensor logits, torch::Tensor labels,
@@ -64,6 +66,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_fwd", &rope_fwd, "RoPE apply fwd/inverse (gfx950)");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward (gfx950)");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward (gfx950)");
+  m.def("gelu_fwd", &gelu_fwd, "exact-erf GELU forward (gfx950)");
+  m.def("gelu_bwd", &gelu_bwd, "exact-erf GELU backward (gfx950)");
   m.def("ce_row_stats", &ce_row_stats, "CE row lse/target (gfx950)");
   m.def("ce_grad_", &ce_grad_, "CE in-place softmax-onehot grad (gfx950)");
   m.def("fused_adamw", &fused_adamw, "multi-tensor AdamW step (gfx950)");

@@ -65,6 +65,54 @@ __global__ void swiglu_bwd_kernel(const T* __restrict__ g, const T* __restrict__
   }
 }
 
+// GELU (K8, pythia MLP — reference modeling_pythia.py:395-406):
+// exact erf form matching F.gelu's default
+DEV_INLINE float gelu_f(float x) {
+  return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+}
+DEV_INLINE float gelu_grad_f(float x) {
+  const float cdf = 0.5f * (1.f + erff(x * 0.70710678118654752f));
+  const float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
+  return cdf + x * pdf;
+}
+
+template <typename T>
+__global__ void gelu_fwd_kernel(const T* __restrict__ x, T* __restrict__ y, long n) {
+  const long vec_n = n / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+       i += (long)gridDim.x * blockDim.x) {
+    Vec8<T> xv = load8(x + i * 8);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = from_f32<T>(gelu_f(to_f32(xv.v[j])));
+    store8(y + i * 8, o);
+  }
+  long start = vec_n * 8;
+  for (long i = start + (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    y[i] = from_f32<T>(gelu_f(to_f32(x[i])));
+}
+
+template <typename T>
+__global__ void gelu_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                T* __restrict__ dx, long n) {
+  const long vec_n = n / 8;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < vec_n;
+       i += (long)gridDim.x * blockDim.x) {
+    Vec8<T> xv = load8(x + i * 8);
+    Vec8<T> dv = load8(dy + i * 8);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o.v[j] = from_f32<T>(to_f32(dv.v[j]) * gelu_grad_f(to_f32(xv.v[j])));
+    store8(dx + i * 8, o);
+  }
+  long start = vec_n * 8;
+  for (long i = start + (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    dx[i] = from_f32<T>(to_f32(dy[i]) * gelu_grad_f(to_f32(x[i])));
+}
+
 static dim3 ew_grid(long n) {
   long blocks = (n / 8 + 255) / 256;
   if (blocks < 1) blocks = 1;
@@ -104,4 +152,36 @@ std::vector<torch::Tensor> swiglu_bwd(torch::Tensor g, torch::Tensor u, torch::T
                        dg.data_ptr<float>(), du.data_ptr<float>(), n);
   HIP_CHECK_LAST();
   return {dg, du};
+}
+
+torch::Tensor gelu_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto y = torch::empty_like(x);
+  const long n = x.numel();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (x.scalar_type() == torch::kBFloat16)
+    hipLaunchKernelGGL(gelu_fwd_kernel<__hip_bfloat16>, ew_grid(n), dim3(256), 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (__hip_bfloat16*)y.data_ptr(), n);
+  else
+    hipLaunchKernelGGL(gelu_fwd_kernel<float>, ew_grid(n), dim3(256), 0, stream,
+                       (const float*)x.data_ptr(), (float*)y.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor gelu_bwd(torch::Tensor x, torch::Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  auto dx = torch::empty_like(x);
+  const long n = x.numel();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  if (x.scalar_type() == torch::kBFloat16)
+    hipLaunchKernelGGL(gelu_bwd_kernel<__hip_bfloat16>, ew_grid(n), dim3(256), 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)dy.data_ptr(),
+                       (__hip_bfloat16*)dx.data_ptr(), n);
+  else
+    hipLaunchKernelGGL(gelu_bwd_kernel<float>, ew_grid(n), dim3(256), 0, stream,
+                       (const float*)x.data_ptr(), (const float*)dy.data_ptr(),
+                       (float*)dx.data_ptr(), n);
+  HIP_CHECK_LAST();
+  return dx;
 }

@@ -201,6 +201,30 @@ def swiglu(gate, up):
 
 
 # ---------------------------------------------------------------------------
+# GELU (K8) — pythia MLP, exact erf form (reference modeling_pythia.py:395-406)
+# ---------------------------------------------------------------------------
+
+
+class _HipGELU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return hip.ext().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return hip.ext().gelu_bwd(x, dy.contiguous())
+
+
+def gelu(x):
+    if hip.use_hip(x, "gelu"):
+        return _HipGELU.apply(x)
+    return F.gelu(x)
+
+
+# ---------------------------------------------------------------------------
 # Flash attention (K3) — causal, no padding mask (parity with the reference:
 # SDPA is_causal=True and padding mask ignored, modeling_llama.py:221-224)
 # ---------------------------------------------------------------------------

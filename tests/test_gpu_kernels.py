@@ -501,3 +501,18 @@ def test_pythia_forward_matches_cpu_hd128():
     for p in gpu.parameters():
         if p.grad is not None:
             assert torch.isfinite(p.grad).all()
+
+
+@pytest.mark.parametrize("n", [4096, 2560 * 3 + 5])
+def test_gelu_fwd_bwd(n):
+    """K8: exact-erf GELU vs F.gelu fp32 (pythia MLP activation)."""
+    torch.manual_seed(0)
+    x = torch.randn(n, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    xr = x.detach().float().requires_grad_(True)
+    y = ext().gelu_fwd(x.detach())
+    ref = F.gelu(xr)
+    assert_close_bf16(y, ref.detach(), atol=1e-2, rtol=1e-2, what="gelu fwd")
+    dy = torch.randn_like(x)
+    dx = ext().gelu_bwd(x.detach(), dy)
+    ref.backward(dy.float())
+    assert_close_bf16(dx, xr.grad, atol=1e-2, rtol=1e-2, what="gelu bwd")
