@@ -1,6 +1,7 @@
 from relora_amd.ops.functional import (  # noqa: F401
     build_rope_cache,
     flash_attention,
+    gelu,
     fused_cross_entropy,
     layernorm,
     lora_linear,
