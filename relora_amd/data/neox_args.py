@@ -67,15 +67,66 @@ class NeoXArgs:
     # everything else from the YAML, kept for introspection
     extra_args: Dict[str, Any] = field(default_factory=dict)
 
+    # type table for the consumed fields — the validation the reference's
+    # dataclass stack applies per-field (arguments.py:109-1240), scoped to
+    # the surface this trainer reads
+    _FIELD_TYPES = {
+        "data_path": str, "train_data_paths": list, "valid_data_paths": list,
+        "test_data_paths": list, "label_data_paths": list,
+        "train_data_weights": list, "valid_data_weights": list,
+        "test_data_weights": list, "weight_by_num_documents": bool,
+        "weighted_sampler_alpha": float, "use_shared_fs": bool,
+        "data_impl": str, "mmap_warmup": bool, "split": str,
+        "seq_length": int, "seed": int, "num_workers": int,
+        "train_iters": int, "eval_iters": int, "eval_interval": int,
+        "iteration": int, "global_num_gpus": int, "train_batch_size": int,
+        "train_micro_batch_size_per_gpu": int,
+        "gradient_accumulation_steps": int, "pipe_parallel_size": int,
+        "model_parallel_size": int, "tokenizer_type": str, "vocab_file": str,
+    }
+
     @classmethod
-    def from_dict(cls, d: Dict[str, Any]):
+    def _check_type(cls, k, v):
+        want = cls._FIELD_TYPES.get(k)
+        if want is None or v is None:
+            return v
+        if want is float and isinstance(v, (int, float)) and not isinstance(v, bool):
+            return float(v)
+        if want is int and isinstance(v, int) and not isinstance(v, bool):
+            return v
+        if want is bool and isinstance(v, bool):
+            return v
+        if want in (str, list) and isinstance(v, want):
+            return v
+        raise ValueError(
+            f"NeoXArgs: field {k!r} expects {want.__name__}, got "
+            f"{type(v).__name__} ({v!r})")
+
+    @classmethod
+    def from_dict(cls, d: Dict[str, Any], strict: Optional[bool] = None):
+        """Build from a YAML dict.  Consumed fields are type-checked; unknown
+        keys are kept in `extra_args` with a warning (they are reference
+        NeoXArgs fields this trainer does not consume), or rejected when
+        `strict=True` / RELORA_AMD_NEOX_STRICT=1 — the fail-on-unknown-key
+        mode of the reference's validation stack."""
+        import os
+        if strict is None:
+            strict = os.environ.get("RELORA_AMD_NEOX_STRICT", "0") == "1"
         known = {f for f in cls.__dataclass_fields__ if f != "extra_args"}
         kwargs, extra = {}, {}
         for k, v in d.items():
             if k in known:
-                kwargs[k] = _none_if_blank(v)
+                kwargs[k] = cls._check_type(k, _none_if_blank(v))
             else:
                 extra[k] = v
+        if extra:
+            if strict:
+                raise ValueError(
+                    f"NeoXArgs(strict): unknown config keys {sorted(extra)}")
+            logger.warning(
+                f"NeoXArgs: {len(extra)} config keys not consumed by this "
+                f"trainer (kept in extra_args): {sorted(extra)[:12]}"
+                f"{' ...' if len(extra) > 12 else ''}")
         args = cls(extra_args=extra, **kwargs)
         args.calculate_derived()
         return args

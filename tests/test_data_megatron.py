@@ -337,3 +337,35 @@ def test_tokenize_and_chunk_offline(tmp_path):
     assert len(rows) == total // 4
     flat = [t for r in rows for t in r]
     assert tok.eos_token_id in flat
+
+
+def test_neox_args_type_validation():
+    """Consumed fields are type-checked (reference arguments.py:109-1240
+    equivalent for the surface this trainer reads)."""
+    import pytest
+    from relora_amd.data.neox_args import NeoXArgs
+
+    base = {"global_num_gpus": 2, "train_micro_batch_size_per_gpu": 4,
+            "data_path": "x", "train_iters": 10}
+    NeoXArgs.from_dict(dict(base))
+    with pytest.raises(ValueError, match="seq_length"):
+        NeoXArgs.from_dict(dict(base, seq_length="2048"))
+    with pytest.raises(ValueError, match="train_data_paths"):
+        NeoXArgs.from_dict(dict(base, train_data_paths="not-a-list"))
+    with pytest.raises(ValueError, match="mmap_warmup"):
+        NeoXArgs.from_dict(dict(base, mmap_warmup=3))
+    # float accepts int
+    a = NeoXArgs.from_dict(dict(base, weighted_sampler_alpha=1))
+    assert a.weighted_sampler_alpha == 1.0
+
+
+def test_neox_args_strict_unknown_keys():
+    import pytest
+    from relora_amd.data.neox_args import NeoXArgs
+
+    base = {"global_num_gpus": 1, "train_micro_batch_size_per_gpu": 2,
+            "data_path": "x", "attention_dropout": 0.1}
+    a = NeoXArgs.from_dict(dict(base))  # non-strict: kept + warned
+    assert a.extra_args == {"attention_dropout": 0.1}
+    with pytest.raises(ValueError, match="unknown config keys"):
+        NeoXArgs.from_dict(dict(base), strict=True)

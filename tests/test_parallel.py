@@ -396,3 +396,80 @@ def test_zero_optimizer_reset_on_shards():
         p.join(timeout=60)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+# ---------------------------------------------------------------------------
+# ZeRO-1 checkpoint cross-compat with torch.distributed ZeroRedundancyOptimizer
+# (reference saves optimizer.pt via ZRO.consolidate_state_dict + state_dict —
+# torchrun_main.py:204-218; both formats are position-indexed torch dicts)
+# ---------------------------------------------------------------------------
+
+
+def _make_model(seed=0):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.Tanh(), torch.nn.Linear(32, 8))
+
+
+def _zero_crosscompat_worker(rank, world, port, q):
+    try:
+        _init(rank, world, port)
+        from torch.distributed.optim import ZeroRedundancyOptimizer
+
+        from relora_amd.parallel.zero import ZeroRedundancyAdamW as ZeroAdamW
+
+        # --- reference-style ZRO produces the checkpoint -------------------
+        ref_model = _make_model()
+        ref_opt = ZeroRedundancyOptimizer(
+            ref_model.parameters(), optimizer_class=torch.optim.AdamW, lr=1e-3)
+        for _ in range(3):
+            ref_model(torch.randn(4, 16)).sum().backward()
+            ref_opt.step()
+            ref_opt.zero_grad()
+        ref_opt.consolidate_state_dict(to=0)
+        sd = ref_opt.state_dict() if rank == 0 else None
+        holder = [sd]
+        torch.distributed.broadcast_object_list(holder, src=0)
+        sd = holder[0]
+
+        # --- our ZeRO-1 loads it -------------------------------------------
+        model = _make_model()
+        opt = ZeroAdamW(model.parameters(), lr=1e-3)
+        opt.load_state_dict(sd)
+        ok = True
+        for i, p in enumerate(opt.all_params):
+            if opt.owner[i] != rank:
+                continue
+            st = opt.optim.state.get(p, {})
+            ok = ok and torch.allclose(st["exp_avg"], sd["state"][i]["exp_avg"])
+            ok = ok and torch.allclose(st["exp_avg_sq"], sd["state"][i]["exp_avg_sq"])
+
+        # --- and round-trips back into a torch ZRO / plain AdamW ------------
+        opt.consolidate_state_dict(to=0)
+        ours = opt.state_dict() if rank == 0 else None
+        if rank == 0:
+            plain = torch.optim.AdamW(_make_model().parameters(), lr=1e-3)
+            plain.load_state_dict(ours)  # torch validates the format
+            for i in sd["state"]:
+                ok = ok and torch.allclose(ours["state"][i]["exp_avg"],
+                                           sd["state"][i]["exp_avg"])
+        q.put((rank, bool(ok)))
+    except Exception as e:  # pragma: no cover
+        import traceback
+        q.put((rank, f"{e}\n{traceback.format_exc()}"))
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_zero1_crosscompat_with_torch_zero_redundancy():
+    world, port = 2, free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_zero_crosscompat_worker, args=(r, world, port, q))
+             for r in range(world)]
+    [p.start() for p in procs]
+    results = [q.get(timeout=180) for _ in range(world)]
+    [p.join(timeout=60) for p in procs]
+    for rank, ok in results:
+        assert ok is True, f"rank {rank}: {ok}"
