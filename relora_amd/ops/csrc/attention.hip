@@ -173,6 +173,60 @@ DEV_INLINE bf16x8 ldsT_frag(const __bf16* tile, int c, int kv0) {
   return *reinterpret_cast<const bf16x8*>(tile + c * TILE + t_rot(kv0 >> 3, c) * 8);
 }
 
+// ---------------------------------------------------------------------------
+// TR layout: gfx950 ds_read_b64_tr_b16 hardware transpose-read (guide T10).
+// The [TILE k-rows][HD cols] operand is stored as [k/4][n/16] subtiles of
+// [4][16] row-major (64 elements, 128 B, contiguous) — the producing stage
+// writes plain b128 rows (no scalar transpose pass, no rotation math), and
+// the MFMA B-fragment materializes via two tr-reads per k-run of 8.
+// Probe-verified semantics (tools/csrc/probe_tr.hip): each 16-lane group's
+// 4-bf16 chunks FLAT[64] redistribute as OUT[lane][j] = FLAT[16j + lane].
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((address_space(3))) const char as3c;
+
+// write a register-staged tile into the TR image: thread's bf16x8 covers
+// (row, c..c+8) -> one b128 store at subtile (row>>2, c>>4) offset
+template <int HD, int NV>
+DEV_INLINE void tile_write_tr(__bf16* dst, const bf16x8 (&r)[NV]) {
+  constexpr int C8 = HD / 8;
+  constexpr int N16 = HD / 16;
+#pragma unroll
+  for (int i = 0; i < NV; ++i) {
+    const int slot = threadIdx.x + i * 512;
+    if (slot >= TILE * C8) break;
+    const int row = slot / C8;
+    const int c = (slot % C8) * 8;
+    const int off = ((row >> 2) * N16 + (c >> 4)) * 64 + (row & 3) * 16 + (c & 15);
+    *reinterpret_cast<bf16x8*>(dst + off) = r[i];
+  }
+}
+
+// B-fragment for the 32x32x16 MFMA from a TR image: lane l holds
+// B[k0 + (l>>5)*8 + j][colbase + (l&31)], j = 0..7
+template <int HD>
+DEV_INLINE bf16x8 tr_frag(const __bf16* tile, int colbase, int k0, int lane) {
+  constexpr int N16 = HD / 16;
+  const int col = colbase + (lane & 31);
+  const int kA = k0 + (lane >> 5) * 8;
+  const int sub1 = (kA >> 2) * N16 + (col >> 4);
+  const int sub2 = sub1 + N16;  // kA+4 row group
+  const int lo = (lane & 15) * 4;
+  as3c* a1 = (as3c*)(tile) + (sub1 * 64 + lo) * 2;
+  as3c* a2 = (as3c*)(tile) + (sub2 * 64 + lo) * 2;
+  unsigned long long r0, r1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(r0), "=v"(r1)
+      : "v"(a1), "v"(a2));
+  union { unsigned long long u[2]; bf16x8 v; } f;
+  f.u[0] = r0;
+  f.u[1] = r1;
+  return f.v;
+}
+
 // write the registered tile transposed into LDS [HD][TILE] (rotated layout)
 template <int HD, int NV>
 DEV_INLINE void tile_write_t(__bf16* dst, const bf16x8 (&r)[NV]) {
@@ -823,7 +877,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
   }
 }
 
-template <int HD>  // HD <= 64 (register budget); hd128 uses the v2 kernel
+template <int HD, bool TRR = false>  // HD <= 64; TRR: tr-read Q^T/dO^T images
 __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
@@ -880,8 +934,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
   tile_load_regs<HD, NV>(rdo, dop, first_qt * TILE, S, hd, ld);
   tile_write_rows<HD, NV>(lds_q, rq, LDK);
   tile_write_rows<HD, NV>(lds_do, rdo, LDK);
-  tile_write_t<HD, NV>(lds_qt, rq);
-  tile_write_t<HD, NV>(lds_dot, rdo);
+  if (TRR) {
+    tile_write_tr<HD, NV>(lds_qt, rq);
+    tile_write_tr<HD, NV>(lds_dot, rdo);
+  } else {
+    tile_write_t<HD, NV>(lds_qt, rq);
+    tile_write_t<HD, NV>(lds_dot, rdo);
+  }
   if (first_qt + 1 < n_q_tiles) {
     tile_load_regs<HD, NV>(rq, qp, (first_qt + 1) * TILE, S, hd, ld);
     tile_load_regs<HD, NV>(rdo, dop, (first_qt + 1) * TILE, S, hd, ld);
@@ -943,8 +1002,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
       if (t == 0 && qt + 1 < n_q_tiles) {
         tile_write_rows<HD, NV>(lds_q + (cur ^ 1) * TILE * LDK, rq, LDK);
         tile_write_rows<HD, NV>(lds_do + (cur ^ 1) * TILE * LDK, rdo, LDK);
-        tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
-        tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+        if (TRR) {
+          tile_write_tr<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+          tile_write_tr<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+        } else {
+          tile_write_t<HD, NV>(lds_qt + (cur ^ 1) * HD * TILE, rq);
+          tile_write_t<HD, NV>(lds_dot + (cur ^ 1) * HD * TILE, rdo);
+        }
         if (qt + 2 < n_q_tiles) {
           tile_load_regs<HD, NV>(rq, qp, (qt + 2) * TILE, S, hd, ld);
           tile_load_regs<HD, NV>(rdo, dop, (qt + 2) * TILE, S, hd, ld);
@@ -959,10 +1023,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dkdv_v3_kernel(
         const int s = 2 * t + g;
 #pragma unroll
         for (int ht = 0; ht < NT32; ++ht) {
-          dv_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pa, ldsT_frag(dotb, ht * 32 + (lane & 31), s * 16 + hi * 8), dv_acc[ht], 0, 0, 0);
-          dk_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dsa, ldsT_frag(qtb, ht * 32 + (lane & 31), s * 16 + hi * 8), dk_acc[ht], 0, 0, 0);
+          const bf16x8 dof = TRR ? tr_frag<HD>(dotb, ht * 32, s * 16, lane)
+                                 : ldsT_frag(dotb, ht * 32 + (lane & 31), s * 16 + hi * 8);
+          const bf16x8 qf2 = TRR ? tr_frag<HD>(qtb, ht * 32, s * 16, lane)
+                                 : ldsT_frag(qtb, ht * 32 + (lane & 31), s * 16 + hi * 8);
+          dv_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dof, dv_acc[ht], 0, 0, 0);
+          dk_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qf2, dk_acc[ht], 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
@@ -1475,14 +1541,27 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
       HIP_CHECK_LAST();
     }
     if (bwd_ver == 3 && HD <= 64) {
+      static const bool trr = [] {
+        const char* e = getenv("RELORA_AMD_ATTN_TR");
+        return e && e[0] == '1';
+      }();
       size_t smem_dkdv3 = (2 * TILE * LDK * 2 + 2 * HD * TILE * 2) * sizeof(__bf16);
-      hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
-                         smem_dkdv3, stream,
-                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
-                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
-                         lse.data_ptr<float>(), delta.data_ptr<float>(),
-                         (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
-                         S, hd, nh, bst, hst, ld, (float)scale);
+      if (trr)
+        hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<HD, true>), dim3((S + 255) / 256, B * nh),
+                           block, smem_dkdv3, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
+                           S, hd, nh, bst, hst, ld, (float)scale);
+      else
+        hipLaunchKernelGGL((attn_bwd_dkdv_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
+                           smem_dkdv3, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(),
+                           S, hd, nh, bst, hst, ld, (float)scale);
       HIP_CHECK_LAST();
     } else {
       constexpr int NBUF = (HD <= 64) ? 2 : 1;
