@@ -219,7 +219,7 @@ DEV_INLINE bf16x8 tr_frag(const __bf16* tile, int colbase, int k0, int lane) {
       "ds_read_b64_tr_b16 %0, %2\n\t"
       "ds_read_b64_tr_b16 %1, %3\n\t"
       "s_waitcnt lgkmcnt(0)"
-      : "=v"(r0), "=v"(r1)
+      : "=&v"(r0), "=&v"(r1)
       : "v"(a1), "v"(a2));
   union { unsigned long long u[2]; bf16x8 v; } f;
   f.u[0] = r0;
@@ -502,7 +502,7 @@ DEV_INLINE bf16x8 pack_p_frag(const float* p16, int g) {
 #define LOG2E 1.44269504088896340736f
 #define DEFER_MAX_THR 11.5f  // log2 domain ~ e^8 (guide T13; bf16 accum headroom)
 
-template <int HD, int MINW = 2>  // MINW: min waves/SIMD hint (4 caps VGPRs at 128)
+template <int HD, int MINW = 2, bool TRR = false>
 __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ out,
@@ -549,8 +549,9 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
   bf16x8 rk[NV], rv[NV];
   tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
   tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
+  if (TRR) tile_write_tr<HD, NV>(lds_vt, rv);
+  else tile_write_t<HD, NV>(lds_vt, rv);
   tile_write_rows<HD, NV>(lds_k, rk, LDK);
-  tile_write_t<HD, NV>(lds_vt, rv);
   if (n_tiles > 1) {
     tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
     tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
@@ -597,7 +598,8 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
       // stage tile kt+1 between the two MFMA clusters (T14 split)
       if (t == 0 && kt + 1 < n_tiles) {
         tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-        tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+        if (TRR) tile_write_tr<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
+        else tile_write_t<HD, NV>(lds_vt + (cur ^ 1) * HD * TILE, rv);
         if (kt + 2 < n_tiles) {
           tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
           tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
@@ -640,9 +642,11 @@ __global__ __launch_bounds__(512, MINW) void attn_fwd_v3_kernel(
         const bf16x8 pfrag = pack_p_frag(p_val, g);
         const int s = 2 * t + g;
 #pragma unroll
-        for (int ht = 0; ht < NT32; ++ht)
-          o_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              ldsT_frag(vtb, ht * 32 + qcol, s * 16 + hi * 8), pfrag, o_acc[ht], 0, 0, 0);
+        for (int ht = 0; ht < NT32; ++ht) {
+          const bf16x8 vf = TRR ? tr_frag<HD>(vtb, ht * 32, s * 16, lane)
+                                : ldsT_frag(vtb, ht * 32 + qcol, s * 16 + hi * 8);
+          o_acc[ht] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pfrag, o_acc[ht], 0, 0, 0);
+        }
       }
       __builtin_amdgcn_s_setprio(0);
     }
@@ -1453,7 +1457,16 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
         const char* e = getenv("RELORA_AMD_ATTN_OCC4");
         return !(e && e[0] == '0');  // default ON (measured +15% at hd64)
       }();
-      if (occ4 && HD <= 64) {
+      static const bool trrf = [] {
+        const char* e = getenv("RELORA_AMD_ATTN_TR");
+        return e && e[0] == '1';
+      }();
+      if (occ4 && HD <= 64 && trrf) {
+        hipLaunchKernelGGL((attn_fwd_v3_kernel<HD, 4, true>), grid, block, smem, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
+                           lse.data_ptr<float>(), S, hd, nh, bst, hst, ld, (float)scale);
+      } else if (occ4 && HD <= 64) {
         hipLaunchKernelGGL((attn_fwd_v3_kernel<HD, 4>), grid, block, smem, stream,
                            (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                            (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)out.data_ptr(),
