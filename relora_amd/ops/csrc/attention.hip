@@ -1458,8 +1458,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k, torch::Ten
         return !(e && e[0] == '0');  // default ON (measured +15% at hd64)
       }();
       static const bool trrf = [] {
-        const char* e = getenv("RELORA_AMD_ATTN_TR");
-        return e && e[0] == '1';
+        const char* e = getenv("RELORA_AMD_ATTN_TR_FWD");
+        return !(e && e[0] == '0');  // default ON: fwd PV tr-read measured +12%
       }();
       if (occ4 && HD <= 64 && trrf) {
         hipLaunchKernelGGL((attn_fwd_v3_kernel<HD, 4, true>), grid, block, smem, stream,
@@ -1555,7 +1555,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
     }
     if (bwd_ver == 3 && HD <= 64) {
       static const bool trr = [] {
-        const char* e = getenv("RELORA_AMD_ATTN_TR");
+        // default OFF: dkdv tr-read measured a wash (155.0 vs 157.1 TF/s) —
+        // the per-fragment lgkmcnt(0) drains offset the staging savings at
+        // 2 waves/SIMD; fwd at 4 waves/SIMD hides them (+12%)
+        const char* e = getenv("RELORA_AMD_ATTN_TR_DKDV");
         return e && e[0] == '1';
       }();
       size_t smem_dkdv3 = (2 * TILE * LDK * 2 + 2 * HD * TILE * 2) * sizeof(__bf16);
