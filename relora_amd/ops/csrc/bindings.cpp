@@ -33,6 +33,7 @@ torch::Tensor multi_tensor_l2norm(std::vector<torch::Tensor> grads);
 void multi_tensor_scale_(std::vector<torch::Tensor> grads, double scale);
 // lora_gemm.hip
 std::vector<torch::Tensor> dropout_mask_fwd(torch::Tensor x, double p, int64_t seed);
+torch::Tensor dropout_mask_bwd(torch::Tensor dy, torch::Tensor mask, double p);
 void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q);
 
 // fused_gemm.hip
@@ -74,6 +75,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "multi-tensor L2 norm (gfx950)");
   m.def("multi_tensor_scale_", &multi_tensor_scale_, "multi-tensor scale (gfx950)");
   m.def("dropout_mask_fwd", &dropout_mask_fwd, "fused dropout + packed mask (gfx950)");
+  m.def("dropout_mask_bwd", &dropout_mask_bwd, "packed-mask dropout backward (gfx950)");
   m.def("lora_add_nt_", &lora_add_nt_, "out += P @ Q^T rank-r MFMA accumulate (gfx950)");
   m.def("fused_lora_gemm", &fused_lora_gemm,
         "y = x@W^T (+bias) + s*t@Bw^T fused MFMA GEMM, 256^2 glds tile (gfx950)");

@@ -169,6 +169,30 @@ __global__ void dropout_mask_kernel(const T* __restrict__ x, T* __restrict__ xd,
   }
 }
 
+// backward of the packed-mask dropout: dx = dy * mask * inv_keep
+template <typename T>
+__global__ void mask_apply_kernel(const T* __restrict__ dy, T* __restrict__ dx,
+                                  const uint8_t* __restrict__ mask, long n8,
+                                  long n, float inv_keep) {
+  const long i8 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i8 >= n8) return;
+  const uint8_t m = mask[i8];
+  if (i8 * 8 + 8 <= n) {
+    Vec8<T> v = load8(dy + i8 * 8);
+    Vec8<T> o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o.v[j] = ((m >> j) & 1) ? from_f32<T>(to_f32(v.v[j]) * inv_keep)
+                              : from_f32<T>(0.f);
+    store8(dx + i8 * 8, o);
+  } else {
+    for (int j = 0; j < 8 && i8 * 8 + j < n; ++j)
+      dx[i8 * 8 + j] = ((m >> j) & 1)
+                           ? from_f32<T>(to_f32(dy[i8 * 8 + j]) * inv_keep)
+                           : from_f32<T>(0.f);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // rank-r accumulate kernel
 // out[M,N] (+)= P[M,r] @ Q^T          (TRANSQ=false: Q is [N,r] row-major)
@@ -606,4 +630,22 @@ torch::Tensor skinny_grad(torch::Tensor P, torch::Tensor X, torch::Tensor xmask,
   }
   HIP_CHECK_LAST();
   return out;
+}
+
+torch::Tensor dropout_mask_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous());
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16);
+  auto dx = torch::empty_like(dy);
+  const long n = dy.numel();
+  const long n8 = (n + 7) / 8;
+  TORCH_CHECK(mask.numel() >= n8, "mask too small");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const float inv_keep = (float)(1.0 / (1.0 - p));
+  hipLaunchKernelGGL(mask_apply_kernel<__hip_bfloat16>,
+                     dim3((n8 + 255) / 256), dim3(256), 0, stream,
+                     (const __hip_bfloat16*)dy.data_ptr(),
+                     (__hip_bfloat16*)dx.data_ptr(),
+                     mask.data_ptr<uint8_t>(), n8, n, inv_keep);
+  HIP_CHECK_LAST();
+  return dx;
 }

@@ -568,6 +568,28 @@ class _QuantizedLoRALinear(torch.autograd.Function):
                 dA, dB, None, None, None, None)
 
 
+class _HipDropout(torch.autograd.Function):
+    """Standalone dropout with the philox packed-bit mask kernels (1 bit per
+    element vs torch's byte mask; the composed-path replacement for
+    F.dropout on odd-dim linears)."""
+
+    @staticmethod
+    def forward(ctx, x, p):
+        xc = x.contiguous()
+        shape = xc.shape
+        seed = _next_dropout_seed() & 0x7FFFFFFFFFFFFFFF
+        xd, mask = hip.ext().dropout_mask_fwd(xc.view(-1, shape[-1]), p, seed)
+        ctx.save_for_backward(mask)
+        ctx.p = p
+        return xd.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        dx = hip.ext().dropout_mask_bwd(dy.contiguous(), mask, ctx.p)
+        return dx.view(dy.shape), None
+
+
 def _fused_ok(x, weight, lora_A, scale, lora_only):
     if weight is None:
         return False
@@ -608,7 +630,13 @@ def lora_linear(x, weight, bias, lora_A, lora_B, scale, dropout_p=0.0,
     if _fused_ok(x, weight, lora_A, scale, lora_only):
         return _FusedLoRALinear.apply(x, weight, bias, lora_A, lora_B,
                                       float(scale), dropout_p, training)
-    xd = F.dropout(x, p=dropout_p, training=training) if dropout_p > 0 else x
+    if dropout_p > 0 and training and x.is_cuda and x.dtype == torch.bfloat16 \
+            and hip.use_hip(x, "dropout"):
+        xd = _HipDropout.apply(x, dropout_p)  # packed 1-bit mask, philox
+    elif dropout_p > 0:
+        xd = F.dropout(x, p=dropout_p, training=training)
+    else:
+        xd = x
     if x.is_cuda:
         # fold the scale into the [M,r] intermediate (45x smaller than the
         # [M,N] product the naive composition scales) — same math, removes

@@ -330,3 +330,26 @@ def test_quantized_relora_no_dense_weight_resident():
     assert not hasattr(m.weight, "weight")
     assert torch.isfinite(m.lora_A.weight.grad).all()
     del y, peak_extra
+
+
+def test_hip_dropout_standalone():
+    """Packed-bit dropout fwd/bwd consistency: backward zeroes exactly the
+    dropped positions and rescales the rest."""
+    from relora_amd.ops.functional import _HipDropout
+
+    torch.manual_seed(0)
+    p = 0.3
+    x = torch.randn(128, 512, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = _HipDropout.apply(x, p)
+    keep = (y != 0)
+    frac = keep.float().mean().item()
+    assert abs(frac - (1 - p)) < 0.03
+    inv = 1.0 / (1 - p)
+    assert torch.allclose(y[keep].float(), x.detach()[keep].float() * inv,
+                          atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(x)
+    y.backward(dy)
+    assert torch.allclose(x.grad[keep].float(), dy[keep].float() * inv,
+                          atol=2e-2, rtol=2e-2)
+    assert (x.grad[~keep] == 0).all()
