@@ -77,9 +77,20 @@ __global__ __launch_bounds__(512) void fused_lora_gemm_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds = (__bf16*)smem;  // [2][A 256x64 | B 256x64] = 128 KiB
 
+  // XCD-aware blockIdx remap (guide T1, bijective form): consecutive
+  // launch ids land on the same XCD's L2 as neighbors in tile space, so
+  // the B tile reused across M-blocks stays resident (+L2 locality on
+  // the K>=2048 shapes)
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  int wgid = blockIdx.x;
+  if (nwg >= 8) {
+    const int xcd = wgid & 7, pos = wgid >> 3;
+    wgid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  }
   const int nbn = (int)(N >> 8);
-  const int bm = blockIdx.x / nbn;
-  const int bn = blockIdx.x % nbn;
+  const int bm = wgid / nbn;
+  const int bn = wgid % nbn;
   const long m0 = (long)bm << 8;
   const long n0 = (long)bn << 8;
 
@@ -258,9 +269,16 @@ __global__ __launch_bounds__(512) void fused_nf4_gemm_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds = (__bf16*)smem;
 
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  int wgid = blockIdx.x;
+  if (nwg >= 8) {
+    const int xcd = wgid & 7, pos = wgid >> 3;
+    wgid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  }
   const int nbn = (int)(N >> 8);
-  const int bm = blockIdx.x / nbn;
-  const int bn = blockIdx.x % nbn;
+  const int bm = wgid / nbn;
+  const int bn = wgid % nbn;
   const long m0 = (long)bm << 8;
   const long n0 = (long)bn << 8;
 
