@@ -52,8 +52,10 @@ def main():
 
     use_gpu = torch.cuda.is_available()
     if use_gpu:
-        torch.cuda.set_device(local_rank)
-        device = f"cuda:{local_rank}"
+        # clamp so a world-2-on-one-GPU proof run (RCCL permitting) works
+        dev = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev)
+        device = f"cuda:{dev}"
         backend = "nccl"  # RCCL on ROCm
     else:
         device = "cpu"
