@@ -39,6 +39,8 @@ void lora_add_nt_(torch::Tensor out, torch::Tensor P, torch::Tensor Q);
 // fused_gemm.hip
 torch::Tensor fused_lora_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor t,
                               torch::Tensor bw, torch::Tensor bias, double lora_scale);
+torch::Tensor fused_lora_gemm3(torch::Tensor x, torch::Tensor w, torch::Tensor t,
+                               torch::Tensor bw, torch::Tensor bias, double lora_scale);
 torch::Tensor fused_nf4_gemm(torch::Tensor x, torch::Tensor qw, torch::Tensor amax,
                              long N, torch::Tensor t, torch::Tensor bw,
                              torch::Tensor bias, double lora_scale);
@@ -79,6 +81,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lora_add_nt_", &lora_add_nt_, "out += P @ Q^T rank-r MFMA accumulate (gfx950)");
   m.def("fused_lora_gemm", &fused_lora_gemm,
         "y = x@W^T (+bias) + s*t@Bw^T fused MFMA GEMM, 256^2 glds tile (gfx950)");
+  m.def("fused_lora_gemm3", &fused_lora_gemm3,
+        "3-buffer counted-vmcnt deep-pipelined fused GEMM variant (gfx950)");
   m.def("fused_nf4_gemm", &fused_nf4_gemm,
         "y = x@dequant(W)^T + s*t@Bw^T with NF4 dequant fused into LDS staging (gfx950)");
   m.def("lora_add_nn_", &lora_add_nn_, "out += maskscale*(P @ Q) rank-r MFMA accumulate (gfx950)");

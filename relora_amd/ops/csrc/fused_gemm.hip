@@ -214,6 +214,182 @@ __global__ __launch_bounds__(512) void fused_lora_gemm_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Deep-pipelined variant: 256x128 tile, THREE LDS buffers, counted vmcnt
+// with raw s_barrier (the guide's "+83% 3-buf span" glds structure — two
+// tiles stay in flight across each barrier; __syncthreads would emit
+// vmcnt(0) and drain them, so the K-loop uses __builtin_amdgcn_s_barrier()
+// + lgkmcnt(0) only).  A[256x64] + B[128x64] = 48 KiB/buffer, 3 buffers =
+// 144 KiB.  8 waves as 2Mx4N -> per-wave 128x32 output (8x2 fragments).
+// Targets the K>=2048 shapes where the 2-buffer kernel loses to hipBLASLt.
+// ---------------------------------------------------------------------------
+
+// stage a [128][64] bf16 tile (16 KB): 16 pieces, 2 per wave
+DEV_INLINE void stage_half_glds(__bf16* image, const __hip_bfloat16* gbase,
+                                long row0, long ld, int k0, int wave, int lane) {
+#pragma unroll
+  for (int p2 = 0; p2 < 2; ++p2) {
+    const int piece = wave * 2 + p2;
+    const unsigned d = piece * 1024u + lane * 16u;
+    const unsigned s = swz(d);
+    const int srow = s >> 7;
+    const int scol_b = s & 127;
+    const __hip_bfloat16* src = gbase + (row0 + srow) * ld + k0 + (scol_b >> 1);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(image) + (piece * 1024u) / 4,
+        16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(512) void fused_lora_gemm3_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    const __hip_bfloat16* __restrict__ t, const __hip_bfloat16* __restrict__ bw,
+    const __hip_bfloat16* __restrict__ bias, __hip_bfloat16* __restrict__ y,
+    long M, long N, long K, int r, float lora_scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = (__bf16*)smem;  // 3 x (A 256x64 + B 128x64) = 144 KiB
+
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  int wgid = blockIdx.x;
+  if (nwg >= 8) {
+    const int xcd = wgid & 7, pos = wgid >> 3;
+    wgid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  }
+  const int nbn = (int)(N >> 7);
+  const int bm = wgid / nbn;
+  const int bn = wgid % nbn;
+  const long m0 = (long)bm << 8;
+  const long n0 = (long)bn << 7;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 2;   // 2 M positions
+  const int wc = wave & 3;    // 4 N positions (32 cols each)
+  const int fr = lane & 15;
+  const int fq = lane >> 4;
+
+  f32x4_g acc[8][2];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4_g{0.f, 0.f, 0.f, 0.f};
+
+  __bf16* bufA[3];
+  __bf16* bufB[3];
+#pragma unroll
+  for (int b = 0; b < 3; ++b) {
+    bufA[b] = lds + b * (256 * 64 + 128 * 64);
+    bufB[b] = bufA[b] + 256 * 64;
+  }
+
+  const int KT = (int)(K >> 6);
+  // A stage = 8 glds/wave (4 pieces x2 halves... 256x64 = 32KB = 4/wave),
+  // B stage = 2/wave; per tile 6 glds/wave.  Keep 2 tiles in flight:
+  // wait vmcnt(12) before computing a tile.
+  stage_tile_glds(bufA[0], x, m0, K, 0, wave, lane);
+  stage_half_glds(bufB[0], w, n0, K, 0, wave, lane);
+  if (KT > 1) {
+    stage_tile_glds(bufA[1], x, m0, K, 64, wave, lane);
+    stage_half_glds(bufB[1], w, n0, K, 64, wave, lane);
+  }
+
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt % 3;
+    if (kt + 2 < KT) {
+      const int nxt = (kt + 2) % 3;
+      stage_tile_glds(bufA[nxt], x, m0, K, (kt + 2) << 6, wave, lane);
+      stage_half_glds(bufB[nxt], w, n0, K, (kt + 2) << 6, wave, lane);
+    }
+    // tile kt's 6 glds are the oldest; up to 12 newer stay in flight
+    if (kt + 2 < KT)
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    else if (kt + 1 < KT)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // raw: no vmcnt(0) drain
+    const __bf16* A = bufA[cur];
+    const __bf16* B = bufB[cur];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_g af[8], bf[2];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        bf[j] = frag_swz(B, wc * 32 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // release buffers for re-staging
+  }
+
+  // LoRA epilogue (serialized 2-buffer staging; r work is ~r/K of the loop)
+  if (r > 0) {
+    const int RC = r >> 6;
+#pragma unroll 1
+    for (int c = 0; c < RC; ++c) {
+      stage_tile_glds(bufA[c & 1], t, m0, r, c << 6, wave, lane);
+      stage_half_glds(bufB[c & 1], bw, n0, r, c << 6, wave, lane);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      const __bf16* A = bufA[c & 1];
+      const __bf16* B = bufB[c & 1];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8_g af[8], bf[2];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          bf[j] = frag_swz(B, wc * 32 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+
+  __syncthreads();
+  // per-wave 128x32 tile through LDS (16 KiB each within the first 128 KiB)
+  __bf16* mine = lds + wave * (128 * 32);
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = i * 16 + fq * 4 + reg;
+        const int col = j * 16 + fr;
+        float v = acc[i][j][reg];
+        if (bias) v += (float)bias[n0 + wc * 32 + col];
+        mine[row * 32 + col] = (__bf16)v;
+      }
+#pragma unroll
+  for (int rr = 0; rr < 8; ++rr) {
+    const int row = rr * 16 + (lane >> 2);
+    const int cb = (lane & 3) * 8;
+    const bf16x8_g vv = *reinterpret_cast<const bf16x8_g*>(mine + row * 32 + cb);
+    *reinterpret_cast<bf16x8_g*>(
+        y + (m0 + wr * 128 + row) * N + n0 + wc * 32 + cb) = vv;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // NF4 variant (K15): the frozen W arrives as packed NF4 (64-element blocks,
 // fp32 absmax, hi-nibble = even element — ops/csrc/quantize.hip layout) and
 // is dequantized DURING LDS staging, so no dense [N,K] W ever exists in HBM
@@ -451,6 +627,40 @@ torch::Tensor fused_lora_gemm(torch::Tensor x, torch::Tensor w, torch::Tensor t,
   dim3 block(512);
   size_t smem = 4 * 256 * 64 * sizeof(__bf16);  // 128 KiB
   hipLaunchKernelGGL(fused_lora_gemm_kernel, grid, block, smem, stream,
+                     (const __hip_bfloat16*)x.data_ptr(),
+                     (const __hip_bfloat16*)w.data_ptr(),
+                     has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
+                     has_lora ? (const __hip_bfloat16*)bw.data_ptr() : nullptr,
+                     (bias.defined() && bias.numel())
+                         ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
+                     (__hip_bfloat16*)y.data_ptr(), M, N, K,
+                     has_lora ? r : 0, (float)lora_scale);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor fused_lora_gemm3(torch::Tensor x, torch::Tensor w, torch::Tensor t,
+                               torch::Tensor bw, torch::Tensor bias, double lora_scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K);
+  TORCH_CHECK(M % 256 == 0 && N % 128 == 0 && K % 64 == 0,
+              "fused_lora_gemm3 requires M%256==0, N%128==0, K%64==0");
+  int r = 0;
+  const bool has_lora = t.defined() && t.numel() > 0;
+  torch::Tensor t_scaled;
+  if (has_lora) {
+    r = (int)t.size(1);
+    TORCH_CHECK(r % 64 == 0 && r <= 128);
+    t_scaled = (lora_scale == 1.0) ? t : (t * lora_scale).contiguous();
+  }
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid((M >> 8) * (N >> 7));
+  dim3 block(512);
+  size_t smem = 3 * (256 * 64 + 128 * 64) * sizeof(__bf16);  // 144 KiB
+  hipLaunchKernelGGL(fused_lora_gemm3_kernel, grid, block, smem, stream,
                      (const __hip_bfloat16*)x.data_ptr(),
                      (const __hip_bfloat16*)w.data_ptr(),
                      has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
