@@ -110,13 +110,13 @@ DEV_INLINE bf16x8 global_frag(const __hip_bfloat16* src, int grow, int S, int hd
 // (512-thread mapping: slot = tid + i*512, row = slot/(HD/8), c8 = slot%(HD/8))
 // ---------------------------------------------------------------------------
 
-template <int HD, int NV>
+template <int HD, int NV, int NT = 512>
 DEV_INLINE void tile_load_regs(bf16x8 (&r)[NV], const __hip_bfloat16* src,
                                int row0, int S, int hd, int ld) {
   constexpr int C8 = HD / 8;
 #pragma unroll
   for (int i = 0; i < NV; ++i) {
-    const int slot = threadIdx.x + i * 512;
+    const int slot = threadIdx.x + i * NT;
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
     const int c = (slot % C8) * 8;
@@ -125,12 +125,12 @@ DEV_INLINE void tile_load_regs(bf16x8 (&r)[NV], const __hip_bfloat16* src,
 }
 
 // write the registered tile row-major into LDS [TILE][ld]
-template <int HD, int NV>
+template <int HD, int NV, int NT = 512>
 DEV_INLINE void tile_write_rows(__bf16* dst, const bf16x8 (&r)[NV], int ld) {
   constexpr int C8 = HD / 8;
 #pragma unroll
   for (int i = 0; i < NV; ++i) {
-    const int slot = threadIdx.x + i * 512;
+    const int slot = threadIdx.x + i * NT;
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
     const int c = (slot % C8) * 8;
@@ -228,12 +228,12 @@ DEV_INLINE bf16x8 tr_frag(const __bf16* tile, int colbase, int k0, int lane) {
 }
 
 // write the registered tile transposed into LDS [HD][TILE] (rotated layout)
-template <int HD, int NV>
+template <int HD, int NV, int NT = 512>
 DEV_INLINE void tile_write_t(__bf16* dst, const bf16x8 (&r)[NV]) {
   constexpr int C8 = HD / 8;
 #pragma unroll
   for (int i = 0; i < NV; ++i) {
-    const int slot = threadIdx.x + i * 512;
+    const int slot = threadIdx.x + i * NT;
     if (slot >= TILE * C8) break;
     const int row = slot / C8;
     const int cb = (slot % C8) * 8;
@@ -739,8 +739,9 @@ DEV_INLINE f32x4 load_f32x4_guard(const float* p, long idx, long n) {
   return r;
 }
 
-template <int HD>
-__global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
+template <int HD, int NWAVE = 8>  // NWAVE*32 q rows per block; 4-wave blocks
+                                  // lift occupancy to 3 blocks/CU at hd64
+__global__ __launch_bounds__(NWAVE * 64) void attn_bwd_dq_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -749,7 +750,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
   constexpr int KSTEPS = HD / 16;
   constexpr int NT32 = HD / 32;
   constexpr int LDK = HD + LPAD;
-  constexpr int NV = (HD + 63) / 64;
+  constexpr int NT = NWAVE * 64;
+  constexpr int NV = (TILE * (HD / 8) + NT - 1) / NT;
+  constexpr int QROWS = NWAVE * 32;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __bf16* lds_k = (__bf16*)smem;             // [2][TILE][LDK] K rows
@@ -757,7 +760,7 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
   __bf16* lds_kt = lds_v + 2 * TILE * LDK;   // [2][HD][TILE]  K^T rotated
 
   const int bh = blockIdx.y;
-  const int q_start = (gridDim.x - 1 - blockIdx.x) * 256;  // heavy blocks first
+  const int q_start = (gridDim.x - 1 - blockIdx.x) * QROWS;  // heavy blocks first
   const long base = (long)(bh / nh) * bst + (long)(bh % nh) * hst;
   const __hip_bfloat16* qp = q + base;
   const __hip_bfloat16* kp = k + base;
@@ -784,18 +787,18 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
 #pragma unroll
   for (int t = 0; t < NT32; ++t) dq_acc[t] = (f32x16)(0.f);
 
-  const int q_max_abs = min(q_start + 255, S - 1);
+  const int q_max_abs = min(q_start + QROWS - 1, S - 1);
   const int n_tiles = (q_max_abs / TILE) + 1;
 
   bf16x8 rk[NV], rv[NV];
-  tile_load_regs<HD, NV>(rk, kp, 0, S, hd, ld);
-  tile_load_regs<HD, NV>(rv, vp, 0, S, hd, ld);
-  tile_write_rows<HD, NV>(lds_k, rk, LDK);
-  tile_write_rows<HD, NV>(lds_v, rv, LDK);
-  tile_write_t<HD, NV>(lds_kt, rk);
+  tile_load_regs<HD, NV, NT>(rk, kp, 0, S, hd, ld);
+  tile_load_regs<HD, NV, NT>(rv, vp, 0, S, hd, ld);
+  tile_write_rows<HD, NV, NT>(lds_k, rk, LDK);
+  tile_write_rows<HD, NV, NT>(lds_v, rv, LDK);
+  tile_write_t<HD, NV, NT>(lds_kt, rk);
   if (n_tiles > 1) {
-    tile_load_regs<HD, NV>(rk, kp, TILE, S, hd, ld);
-    tile_load_regs<HD, NV>(rv, vp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV, NT>(rk, kp, TILE, S, hd, ld);
+    tile_load_regs<HD, NV, NT>(rv, vp, TILE, S, hd, ld);
   }
   __syncthreads();
 
@@ -841,12 +844,12 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_v3_kernel(
       __builtin_amdgcn_s_setprio(0);
 
       if (t == 0 && kt + 1 < n_tiles) {
-        tile_write_rows<HD, NV>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
-        tile_write_rows<HD, NV>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
-        tile_write_t<HD, NV>(lds_kt + (cur ^ 1) * HD * TILE, rk);
+        tile_write_rows<HD, NV, NT>(lds_k + (cur ^ 1) * TILE * LDK, rk, LDK);
+        tile_write_rows<HD, NV, NT>(lds_v + (cur ^ 1) * TILE * LDK, rv, LDK);
+        tile_write_t<HD, NV, NT>(lds_kt + (cur ^ 1) * HD * TILE, rk);
         if (kt + 2 < n_tiles) {
-          tile_load_regs<HD, NV>(rk, kp, (kt + 2) * TILE, S, hd, ld);
-          tile_load_regs<HD, NV>(rv, vp, (kt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV, NT>(rk, kp, (kt + 2) * TILE, S, hd, ld);
+          tile_load_regs<HD, NV, NT>(rv, vp, (kt + 2) * TILE, S, hd, ld);
         }
       }
 
@@ -1534,13 +1537,27 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
     const int LDK = HD + LPAD, LDT = TILE + LPAD;
     if (bwd_ver == 3 && HD <= 96) {  // dq v3 at hd128 hits the VGPR cap
       size_t smem_dq3 = (2 * TILE * LDK * 2 + 2 * HD * TILE) * sizeof(__bf16);
-      hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
-                         smem_dq3, stream,
-                         (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
-                         (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
-                         lse.data_ptr<float>(), delta.data_ptr<float>(),
-                         (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
-                         (float)scale);
+      static const bool dq4w = [] {
+        // 4-wave blocks: 158-VGPR kernel fits 3 blocks/CU (vs 2 waves/SIMD)
+        const char* e = getenv("RELORA_AMD_DQ_4WAVE");
+        return e && e[0] == '1';
+      }();
+      if (dq4w && HD <= 64)
+        hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD, 4>), dim3((S + 127) / 128, B * nh),
+                           dim3(256), smem_dq3, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
+                           (float)scale);
+      else
+        hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD>), dim3((S + 255) / 256, B * nh), block,
+                           smem_dq3, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
+                           (float)scale);
       HIP_CHECK_LAST();
     } else {
       size_t smem_dq = (2 * TILE * LDK * 2 + 2 * HD * TILE + 8 * 16 * LDT) * sizeof(__bf16);
