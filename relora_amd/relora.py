@@ -283,7 +283,22 @@ class ReLoRaLinear(nn.Module):
             logger.warning("Skipping merge and reinit, because only lora parameters are used")
             return
         if self.quantize is None:
-            self.weight.data += self.lora_B.weight @ self.lora_A.weight * self._post_lora_scale()
+            merged = False
+            if (self.weight.is_cuda and self.weight.dtype == torch.bfloat16
+                    and self.r % 32 == 0 and self.r <= 256
+                    and not torch.is_tensor(self._post_lora_scale())):
+                from relora_amd.ops import hip as _hip
+                if _hip.use_hip(self.weight, "merge"):
+                    # K13 on the MFMA accumulate kernel: W += (B*s) @ A
+                    # in-place (fp32 accumulation, single bf16 rounding —
+                    # the torch form rounds the product before the add)
+                    bs = (self.lora_B.weight * self._post_lora_scale()).contiguous()
+                    _hip.ext().lora_add_nn_(
+                        self.weight.data, bs, self.lora_A.weight.contiguous(),
+                        bs.new_empty(0, dtype=torch.uint8), 1.0)
+                    merged = True
+            if not merged:
+                self.weight.data += self.lora_B.weight @ self.lora_A.weight * self._post_lora_scale()
         else:
             # dequant -> merge -> requant (reference relora.py:277-299)
             w = self.weight.materialize(self.lora_A.weight.dtype)

@@ -373,3 +373,30 @@ def test_fused_lora_gemm_variants_match_ref():
         err = (y.float() - ref).abs()
         tol = 2e-2 + 2e-2 * ref.abs().clamp_min(1.0)
         assert (err <= tol).all(), f"{fn}: {err.max().item()}"
+
+
+def test_merge_and_reinit_gpu_kernel_path():
+    """K13: merge runs on the MFMA accumulate kernel on GPU; result matches
+    the torch composition within bf16 rounding."""
+    from relora_amd.relora import ReLoRaLinear
+
+    torch.manual_seed(0)
+    for in_f, out_f in [(256, 512), (2048, 5461)]:  # aligned + odd out-dim
+        lin = torch.nn.Linear(in_f, out_f, bias=False)
+        m = ReLoRaLinear(in_f, out_f, r=128, lora_alpha=32, lora_dropout=0.0,
+                         weight_data=lin.weight.data.clone(), bias_data=None)
+        m = m.to("cuda", torch.bfloat16)
+        with torch.no_grad():
+            m.lora_B.weight.normal_(0, 0.02)  # nonzero so the merge does work
+        w_before = m.weight.data.clone()
+        ref = (w_before.float()
+               + (m.lora_B.weight.float() @ m.lora_A.weight.float())
+               * m._post_lora_scale())
+        a_before = m.lora_A.weight.clone()
+        m.merge_and_reinit()
+        err = (m.weight.data.float() - ref).abs()
+        tol = 2e-2 + 2e-2 * ref.abs().clamp_min(1.0)
+        assert (err <= tol).all(), err.max()
+        # reinit happened
+        assert (m.lora_B.weight == 0).all()
+        assert not torch.equal(m.lora_A.weight, a_before)
