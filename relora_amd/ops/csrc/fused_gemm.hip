@@ -390,6 +390,197 @@ __global__ __launch_bounds__(512) void fused_lora_gemm3_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// BK=32 / 4-buffer variant: the guide's "3-buffer glds + raw barrier" tier
+// (measured 1156-1182 TF at 4k³ there) — counted vmcnt keeps THREE
+// sub-tiles of glds in flight across every barrier, so the per-tile
+// vmcnt(0) drain that caps the 2-buffer kernel at ~820 TF disappears.
+// A[256][32] + B[256][32] = 32 KiB per buffer x 4 = 128 KiB.
+// Rows are 64 B here, so the st_16x32 swizzle does not apply; instead the
+// 16-B slot rotates by the HIGH row bits (slot ^= (row>>2)&3), which makes
+// each 16-lane ds_read_b128 group tile all 64 banks exactly (conflict-free;
+// derivation in the comment above swz32).
+// ---------------------------------------------------------------------------
+
+// swizzle for [rows][32] bf16 images: byte bit4..5 ^= row bits 2..3
+DEV_INLINE unsigned swz32(unsigned o) {
+  return o ^ ((((o >> 8) & 3u)) << 4);  // row = o>>6; (row>>2)&3 = (o>>8)&3
+}
+
+// stage a [256][32] bf16 sub-tile (16 KB): 16 pieces, 2 per wave
+DEV_INLINE void stage_k32_glds(__bf16* image, const __hip_bfloat16* gbase,
+                               long row0, long ld, int k0, int wave, int lane) {
+#pragma unroll
+  for (int p2 = 0; p2 < 2; ++p2) {
+    const int piece = wave * 2 + p2;
+    const unsigned d = piece * 1024u + lane * 16u;
+    const unsigned s = swz32(d);
+    const int srow = s >> 6;       // 64-B rows
+    const int scol_b = s & 63;
+    const __hip_bfloat16* src = gbase + (row0 + srow) * ld + k0 + (scol_b >> 1);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(image) + (piece * 1024u) / 4,
+        16, 0, 0);
+  }
+}
+
+DEV_INLINE bf16x8_g frag_k32(const __bf16* image, int row) {
+  const unsigned o = (unsigned)row * 64u + 0u;  // k-offset folded by caller
+  return *reinterpret_cast<const bf16x8_g*>((const char*)image + swz32(o));
+}
+
+// fragment at (row, k-quarter fq*8): logical byte = row*64 + fq*16
+DEV_INLINE bf16x8_g frag_k32q(const __bf16* image, int row, int fq) {
+  const unsigned o = (unsigned)row * 64u + (unsigned)fq * 16u;
+  return *reinterpret_cast<const bf16x8_g*>((const char*)image + swz32(o));
+}
+
+__global__ __launch_bounds__(512) void fused_lora_gemm4_kernel(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    const __hip_bfloat16* __restrict__ t, const __hip_bfloat16* __restrict__ bw,
+    const __hip_bfloat16* __restrict__ bias, __hip_bfloat16* __restrict__ y,
+    long M, long N, long K, int r) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = (__bf16*)smem;  // 4 x (A 256x32 + B 256x32) = 128 KiB
+
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  int wgid = blockIdx.x;
+  if (nwg >= 8) {
+    const int xcd = wgid & 7, pos = wgid >> 3;
+    wgid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  }
+  const int nbn = (int)(N >> 8);
+  const int bm = wgid / nbn;
+  const int bn = wgid % nbn;
+  const long m0 = (long)bm << 8;
+  const long n0 = (long)bn << 8;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  f32x4_g acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4_g{0.f, 0.f, 0.f, 0.f};
+
+  __bf16* bufA[4];
+  __bf16* bufB[4];
+#pragma unroll
+  for (int b = 0; b < 4; ++b) {
+    bufA[b] = lds + b * (2 * 256 * 32);
+    bufB[b] = bufA[b] + 256 * 32;
+  }
+
+  const int ST = (int)(K >> 5);  // 32-deep sub-tiles
+  // 4 glds per wave per sub-tile (2 A + 2 B); 3 sub-tiles stay in flight
+  stage_k32_glds(bufA[0], x, m0, K, 0, wave, lane);
+  stage_k32_glds(bufB[0], w, n0, K, 0, wave, lane);
+  if (ST > 1) {
+    stage_k32_glds(bufA[1], x, m0, K, 32, wave, lane);
+    stage_k32_glds(bufB[1], w, n0, K, 32, wave, lane);
+  }
+  if (ST > 2) {
+    stage_k32_glds(bufA[2], x, m0, K, 64, wave, lane);
+    stage_k32_glds(bufB[2], w, n0, K, 64, wave, lane);
+  }
+
+  for (int s = 0; s < ST; ++s) {
+    const int cur = s & 3;
+    if (s + 3 < ST) {
+      const int nxt = (s + 3) & 3;
+      stage_k32_glds(bufA[nxt], x, m0, K, (s + 3) << 5, wave, lane);
+      stage_k32_glds(bufB[nxt], w, n0, K, (s + 3) << 5, wave, lane);
+    }
+    // leave every sub-tile newer than s in flight
+    const int ahead = min(3, ST - 1 - s);
+    if (ahead >= 3)
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    else if (ahead == 2)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else if (ahead == 1)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // raw: glds stay in flight
+    asm volatile("" ::: "memory");  // compiler fence: no read hoists above
+    const __bf16* A = bufA[cur];
+    const __bf16* B = bufB[cur];
+    bf16x8_g af[8], bf[4];
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      af[i] = frag_k32q(A, wr * 128 + i * 16 + fr, fq);
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      bf[j] = frag_k32q(B, wc * 64 + j * 16 + fr, fq);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bf[j], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // buffer release
+  }
+
+  // LoRA epilogue: reuse the 2-buffer serialized pattern over 32-deep chunks
+  if (r > 0) {
+    const int RC = r >> 5;
+#pragma unroll 1
+    for (int c = 0; c < RC; ++c) {
+      stage_k32_glds(bufA[c & 1], t, m0, r, c << 5, wave, lane);
+      stage_k32_glds(bufB[c & 1], bw, n0, r, c << 5, wave, lane);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      const __bf16* A = bufA[c & 1];
+      const __bf16* B = bufB[c & 1];
+      bf16x8_g af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = frag_k32q(A, wr * 128 + i * 16 + fr, fq);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_k32q(B, wc * 64 + j * 16 + fr, fq);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+      __syncthreads();
+    }
+  }
+
+  __syncthreads();
+  __bf16* mine = lds + wave * (128 * 64);
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = i * 16 + fq * 4 + reg;
+        const int col = j * 16 + fr;
+        float v = acc[i][j][reg];
+        if (bias) v += (float)bias[n0 + wc * 64 + col];
+        mine[row * 64 + col] = (__bf16)v;
+      }
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    const int row = rr * 8 + (lane >> 3);
+    const int cb = (lane & 7) * 8;
+    const bf16x8_g vv = *reinterpret_cast<const bf16x8_g*>(mine + row * 64 + cb);
+    *reinterpret_cast<bf16x8_g*>(
+        y + (m0 + wr * 128 + row) * N + n0 + wc * 64 + cb) = vv;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // NF4 variant (K15): the frozen W arrives as packed NF4 (64-element blocks,
 // fp32 absmax, hi-nibble = even element — ops/csrc/quantize.hip layout) and
 // is dequantized DURING LDS staging, so no dense [N,K] W ever exists in HBM
@@ -669,6 +860,39 @@ torch::Tensor fused_lora_gemm3(torch::Tensor x, torch::Tensor w, torch::Tensor t
                          ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
                      (__hip_bfloat16*)y.data_ptr(), M, N, K,
                      has_lora ? r : 0, (float)lora_scale);
+  HIP_CHECK_LAST();
+  return y;
+}
+
+torch::Tensor fused_lora_gemm4(torch::Tensor x, torch::Tensor w, torch::Tensor t,
+                               torch::Tensor bw, torch::Tensor bias, double lora_scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K);
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 32 == 0,
+              "fused_lora_gemm4 requires M%256==0, N%256==0, K%32==0");
+  int r = 0;
+  const bool has_lora = t.defined() && t.numel() > 0;
+  torch::Tensor t_scaled;
+  if (has_lora) {
+    r = (int)t.size(1);
+    TORCH_CHECK(r % 32 == 0 && r <= 256);
+    t_scaled = (lora_scale == 1.0) ? t : (t * lora_scale).contiguous();
+  }
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid((M >> 8) * (N >> 8));
+  dim3 block(512);
+  size_t smem = 4 * 2 * 256 * 32 * sizeof(__bf16);  // 128 KiB
+  hipLaunchKernelGGL(fused_lora_gemm4_kernel, grid, block, smem, stream,
+                     (const __hip_bfloat16*)x.data_ptr(),
+                     (const __hip_bfloat16*)w.data_ptr(),
+                     has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
+                     has_lora ? (const __hip_bfloat16*)bw.data_ptr() : nullptr,
+                     (bias.defined() && bias.numel())
+                         ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
+                     (__hip_bfloat16*)y.data_ptr(), M, N, K, has_lora ? r : 0);
   HIP_CHECK_LAST();
   return y;
 }

@@ -81,11 +81,18 @@ def main():
                    for _ in range(4))
         t3 = timeit(lambda: ext.fused_lora_gemm3(x, w, t, bw, empty, scale),
                     args.iters)
+        y4a = ext.fused_lora_gemm4(x, w, t, bw, empty, scale)
+        bad4 = ((y4a.float() - y_ref).abs() > tol).sum().item()
+        det4 = sum((ext.fused_lora_gemm4(x, w, t, bw, empty, scale) != y4a).sum().item()
+                   for _ in range(4))
+        t4 = timeit(lambda: ext.fused_lora_gemm4(x, w, t, bw, empty, scale),
+                    args.iters)
         print(f"{name:12s} M{M} N{N} K{K} r{r}: "
               f"composed {t_comp*1e6:7.1f} us ({flops/t_comp/1e12:6.0f} TF) | "
               f"fused {t_fused*1e6:7.1f} us ({flops/t_fused/1e12:6.0f} TF) | "
               f"3buf {t3*1e6:7.1f} us ({flops/t3/1e12:6.0f} TF) | "
-              f"bad={bad}/{bad3} nondet3={det3} "
+              f"bk32x4 {t4*1e6:7.1f} us ({flops/t4/1e12:6.0f} TF) | "
+              f"bad={bad}/{bad3}/{bad4} nondet={det3}/{det4} "
               f"maxerr={err.max().item():.4f}")
 
 
