@@ -424,11 +424,6 @@ DEV_INLINE void stage_k32_glds(__bf16* image, const __hip_bfloat16* gbase,
   }
 }
 
-DEV_INLINE bf16x8_g frag_k32(const __bf16* image, int row) {
-  const unsigned o = (unsigned)row * 64u + 0u;  // k-offset folded by caller
-  return *reinterpret_cast<const bf16x8_g*>((const char*)image + swz32(o));
-}
-
 // fragment at (row, k-quarter fq*8): logical byte = row*64 + fq*16
 DEV_INLINE bf16x8_g frag_k32q(const __bf16* image, int row, int fq) {
   const unsigned o = (unsigned)row * 64u + (unsigned)fq * 16u;
