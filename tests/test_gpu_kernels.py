@@ -516,3 +516,54 @@ def test_gelu_fwd_bwd(n):
     dx = ext().gelu_bwd(x.detach(), dy)
     ref.backward(dy.float())
     assert_close_bf16(dx, xr.grad, atol=1e-2, rtol=1e-2, what="gelu bwd")
+
+
+def test_attention_zero_copy_layouts():
+    """The q/k/v path must stay copy-free: flash_attention accepts the
+    model's transposed-BSHD views and returns the SAME layout, so the
+    model's .transpose(1,2).reshape is a free view (regression guard for
+    the strided-attention property)."""
+    from relora_amd import ops as fops
+
+    B, S, nh, hd = 2, 256, 4, 64
+    x = torch.randn(B, S, nh, hd, device="cuda", dtype=torch.bfloat16)
+    q = x.view(B, S, nh, hd).transpose(1, 2)
+    k = torch.randn_like(x).transpose(1, 2)
+    v = torch.randn_like(x).transpose(1, 2)
+    assert not q.is_contiguous()  # transposed view, no copy
+    o = fops.flash_attention(q, k, v, causal=True)
+    assert o.stride() == q.stride(), "output layout must match input views"
+    # the model's epilogue view chain is then free
+    o2 = o.transpose(1, 2)
+    assert o2.is_contiguous()
+    # rope keeps the layout too
+    cos, sin = fops.build_rope_cache(hd, S, device="cuda")
+    qo, ko = fops.rope(q, k, cos, sin)
+    assert qo.stride() == q.stride()
+
+
+def test_pythia_train_smoke_gpu():
+    """Second model family end-to-end on the GPU kernel stack: a few
+    ReLoRA-wrapped pythia steps descend and stay finite."""
+    from relora_amd.models import build_model_from_config, load_model_config
+    from relora_amd.ops.optim import AdamW
+    from relora_amd.relora import ReLoRaModel
+
+    torch.manual_seed(0)
+    cfg = load_model_config("configs/pythia_160m.json")
+    model = build_model_from_config(cfg)
+    model = ReLoRaModel(model, r=32, lora_alpha=32, lora_dropout=0.1,
+                        target_modules=["attn", "attention", "mlp"],
+                        keep_original_weights=True)
+    model = model.to("cuda", torch.bfloat16).train()
+    opt = AdamW([p for p in model.parameters() if p.requires_grad], lr=3e-4)
+    x = torch.randint(0, cfg.vocab_size, (2, 256), device="cuda")
+    losses = []
+    for _ in range(8):
+        loss = model(input_ids=x, labels=x).loss
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] - 0.5, losses
