@@ -43,6 +43,9 @@ torch::Tensor fused_lora_gemm3(torch::Tensor x, torch::Tensor w, torch::Tensor t
                                torch::Tensor bw, torch::Tensor bias, double lora_scale);
 torch::Tensor fused_lora_gemm4(torch::Tensor x, torch::Tensor w, torch::Tensor t,
                                torch::Tensor bw, torch::Tensor bias, double lora_scale);
+torch::Tensor fused_int8_gemm(torch::Tensor x, torch::Tensor qw, torch::Tensor amax,
+                              long N, torch::Tensor t, torch::Tensor bw,
+                              torch::Tensor bias, double lora_scale);
 torch::Tensor fused_nf4_gemm(torch::Tensor x, torch::Tensor qw, torch::Tensor amax,
                              long N, torch::Tensor t, torch::Tensor bw,
                              torch::Tensor bias, double lora_scale);
@@ -87,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "3-buffer counted-vmcnt deep-pipelined fused GEMM variant (gfx950)");
   m.def("fused_lora_gemm4", &fused_lora_gemm4,
         "BK=32 4-buffer counted-vmcnt fused GEMM (glds-span tier, gfx950)");
+  m.def("fused_int8_gemm", &fused_int8_gemm,
+        "y = x@dequant_int8(W)^T + s*t@Bw^T, dequant fused into staging (gfx950)");
   m.def("fused_nf4_gemm", &fused_nf4_gemm,
         "y = x@dequant(W)^T + s*t@Bw^T with NF4 dequant fused into LDS staging (gfx950)");
   m.def("lora_add_nn_", &lora_add_nn_, "out += maskscale*(P @ Q) rank-r MFMA accumulate (gfx950)");

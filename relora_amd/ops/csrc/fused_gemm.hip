@@ -575,6 +575,180 @@ __global__ __launch_bounds__(512) void fused_lora_gemm4_kernel(
   }
 }
 
+
+// int8 variant of the dequant stage (256-element symmetric blocks,
+// scale = absmax/127 — ops/csrc/quantize.hip layout).  Requires K%256==0
+// so each thread's 32-element run stays inside one block.
+DEV_INLINE void stage_tile_int8(__bf16* image, const int8_t* qdata,
+                                const float* absmax, long row0, long K, int k0,
+                                int tid) {
+  const int row = tid >> 1;
+  const int half = tid & 1;
+  const long flat = (row0 + row) * K + k0 + half * 32;
+  const float sc = absmax[flat >> 8] * (1.f / 127.f);
+  union { uint32_t u[8]; int8_t b[32]; } pk;
+  *reinterpret_cast<uint4*>(pk.u) = *reinterpret_cast<const uint4*>(qdata + flat);
+  *reinterpret_cast<uint4*>(pk.u + 4) = *reinterpret_cast<const uint4*>(qdata + flat + 16);
+  __bf16 vals[32];
+#pragma unroll
+  for (int i = 0; i < 32; ++i) vals[i] = (__bf16)(pk.b[i] * sc);
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const unsigned o = (unsigned)row * 128u + (unsigned)half * 64u + c * 16u;
+    *reinterpret_cast<bf16x8_g*>((char*)image + swz(o)) =
+        *reinterpret_cast<const bf16x8_g*>(vals + c * 8);
+  }
+}
+
+__global__ __launch_bounds__(512) void fused_int8_gemm_kernel(
+    const __hip_bfloat16* __restrict__ x, const int8_t* __restrict__ qw,
+    const float* __restrict__ amax, const __hip_bfloat16* __restrict__ t,
+    const __hip_bfloat16* __restrict__ bw, const __hip_bfloat16* __restrict__ bias,
+    __hip_bfloat16* __restrict__ y, long M, long N, long K, int r) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* lds = (__bf16*)smem;
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  int wgid = blockIdx.x;
+  if (nwg >= 8) {
+    const int xcd = wgid & 7, pos = wgid >> 3;
+    wgid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + pos;
+  }
+  const int nbn = (int)(N >> 8);
+  const long m0 = (long)(wgid / nbn) << 8;
+  const long n0 = (long)(wgid % nbn) << 8;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int wr = wave >> 2, wc = wave & 3;
+  const int fr = lane & 15, fq = lane >> 4;
+  f32x4_g acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4_g{0.f, 0.f, 0.f, 0.f};
+  __bf16* bufA[2] = {lds, lds + 2 * 256 * 64};
+  __bf16* bufB[2] = {lds + 256 * 64, lds + 3 * 256 * 64};
+  const int KT = (int)(K >> 6);
+  stage_tile_glds(bufA[0], x, m0, K, 0, wave, lane);
+  stage_tile_int8(bufB[0], qw, amax, n0, K, 0, tid);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  for (int kt = 0; kt < KT; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < KT) {
+      stage_tile_glds(bufA[cur ^ 1], x, m0, K, (kt + 1) << 6, wave, lane);
+      stage_tile_int8(bufB[cur ^ 1], qw, amax, n0, K, (kt + 1) << 6, tid);
+    }
+    const __bf16* A = bufA[cur];
+    const __bf16* B = bufB[cur];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_g af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+  }
+  if (r > 0) {
+    const int RC = r >> 6;
+#pragma unroll 1
+    for (int c = 0; c < RC; ++c) {
+      stage_tile_glds(bufA[c & 1], t, m0, r, c << 6, wave, lane);
+      stage_tile_glds(bufB[c & 1], bw, n0, r, c << 6, wave, lane);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      const __bf16* A = bufA[c & 1];
+      const __bf16* B = bufB[c & 1];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8_g af[8], bf[4];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          af[i] = frag_swz(A, wr * 128 + i * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          bf[j] = frag_swz(B, wc * 64 + j * 16 + fr, ks * 32 + fq * 8);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+  __syncthreads();
+  __bf16* mine = lds + wave * (128 * 64);
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int row = i * 16 + fq * 4 + reg;
+        const int col = j * 16 + fr;
+        float v = acc[i][j][reg];
+        if (bias) v += (float)bias[n0 + wc * 64 + col];
+        mine[row * 64 + col] = (__bf16)v;
+      }
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    const int row = rr * 8 + (lane >> 3);
+    const int cb = (lane & 7) * 8;
+    const bf16x8_g vv = *reinterpret_cast<const bf16x8_g*>(mine + row * 64 + cb);
+    *reinterpret_cast<bf16x8_g*>(
+        y + (m0 + wr * 128 + row) * N + n0 + wc * 64 + cb) = vv;
+  }
+}
+
+torch::Tensor fused_int8_gemm(torch::Tensor x, torch::Tensor qw, torch::Tensor amax,
+                              long N, torch::Tensor t, torch::Tensor bw,
+                              torch::Tensor bias, double lora_scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(qw.scalar_type() == torch::kInt8 || qw.scalar_type() == torch::kChar);
+  const long M = x.size(0), K = x.size(1);
+  TORCH_CHECK(qw.numel() == N * K, "packed int8 size mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 256 == 0,
+              "fused_int8_gemm requires M%256==0, N%256==0, K%256==0");
+  int r = 0;
+  const bool has_lora = t.defined() && t.numel() > 0;
+  torch::Tensor t_scaled;
+  if (has_lora) {
+    r = (int)t.size(1);
+    TORCH_CHECK(r % 64 == 0 && r <= 256);
+    t_scaled = (lora_scale == 1.0) ? t : (t * lora_scale).contiguous();
+  }
+  auto y = torch::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  dim3 grid((M >> 8) * (N >> 8));
+  dim3 block(512);
+  size_t smem = 4 * 256 * 64 * sizeof(__bf16);
+  hipLaunchKernelGGL(fused_int8_gemm_kernel, grid, block, smem, stream,
+                     (const __hip_bfloat16*)x.data_ptr(), (const int8_t*)qw.data_ptr(),
+                     amax.data_ptr<float>(),
+                     has_lora ? (const __hip_bfloat16*)t_scaled.data_ptr() : nullptr,
+                     has_lora ? (const __hip_bfloat16*)bw.data_ptr() : nullptr,
+                     (bias.defined() && bias.numel())
+                         ? (const __hip_bfloat16*)bias.data_ptr() : nullptr,
+                     (__hip_bfloat16*)y.data_ptr(), M, N, K, has_lora ? r : 0);
+  HIP_CHECK_LAST();
+  return y;
+}
+
 // ---------------------------------------------------------------------------
 // NF4 variant (K15): the frozen W arrives as packed NF4 (64-element blocks,
 // fp32 absmax, hi-nibble = even element — ops/csrc/quantize.hip layout) and

@@ -524,11 +524,16 @@ class _QuantizedLoRALinear(torch.autograd.Function):
         del xd
         r = lora_A.shape[0]
         empty = x2d.new_empty(0)
-        fused_ok = (qw.mode == "4bit" and M % 256 == 0 and N % 256 == 0
-                    and K % 64 == 0 and r % 64 == 0 and r <= 256
-                    and _FUSED_K1 != "0")
-        if fused_ok:
+        aligned = (M % 256 == 0 and N % 256 == 0 and r % 64 == 0
+                   and r <= 256 and _FUSED_K1 != "0")
+        fused_nf4 = aligned and qw.mode == "4bit" and K % 64 == 0
+        fused_i8 = aligned and qw.mode == "8bit" and K % 256 == 0
+        if fused_nf4:
             y = hip.ext().fused_nf4_gemm(
+                x2d, qw.qdata, qw.absmax, N, t_u, lora_B,
+                bias if bias is not None else empty, scale)
+        elif fused_i8:
+            y = hip.ext().fused_int8_gemm(
                 x2d, qw.qdata, qw.absmax, N, t_u, lora_B,
                 bias if bias is not None else empty, scale)
         else:

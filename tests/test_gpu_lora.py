@@ -400,3 +400,20 @@ def test_merge_and_reinit_gpu_kernel_path():
         # reinit happened
         assert (m.lora_B.weight == 0).all()
         assert not torch.equal(m.lora_A.weight, a_before)
+
+
+def test_fused_int8_gemm_matches_dequant_matmul():
+    """K15 (8bit): dequant-fused int8 GEMM vs dequantize-then-matmul."""
+    torch.manual_seed(0)
+    M, N, K, r = 512, 256, 256, 64
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    t = torch.randn(M, r, device="cuda", dtype=torch.bfloat16) * 0.1
+    bw = torch.randn(N, r, device="cuda", dtype=torch.bfloat16) * 0.1
+    q, am = ext().quantize_int8(w.reshape(-1).contiguous())
+    wd = ext().dequantize_int8(q, am, w.numel(), torch.bfloat16).view(N, K)
+    scale = 0.5
+    ref = (x.float() @ wd.float().t()) + scale * (t.float() @ bw.float().t())
+    got = ext().fused_int8_gemm(x, q, am, N, t, bw, x.new_empty(0), scale)
+    err = (got.float() - ref).abs()
+    assert err.max().item() < 2e-2 + 2e-2 * ref.abs().max().item(), err.max()
