@@ -157,11 +157,13 @@ class LlamaDecoderLayer(nn.Module):
             hidden_states, position_ids=position_ids,
             past_key_value=past_key_value, use_cache=use_cache,
         )
-        hidden_states = residual + hidden_states
-
-        residual = hidden_states
-        hidden_states = self.post_attention_layernorm(hidden_states)
-        hidden_states = self.mlp(hidden_states)
+        # fused residual add + norm (K16): same bf16 rounding as the
+        # unfused `residual + h` followed by rmsnorm; one kernel fewer in
+        # each direction on GPU
+        normed, residual = ops.add_rmsnorm(
+            hidden_states, residual, self.post_attention_layernorm.weight,
+            self.post_attention_layernorm.variance_epsilon)
+        hidden_states = self.mlp(normed)
         hidden_states = residual + hidden_states
         return hidden_states, present
 

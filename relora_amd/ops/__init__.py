@@ -5,6 +5,7 @@ from relora_amd.ops.functional import (  # noqa: F401
     fused_cross_entropy,
     layernorm,
     lora_linear,
+    add_rmsnorm,
     rmsnorm,
     rmsnorm_torch,
     rope,

@@ -4,6 +4,11 @@
 
 // norms.hip
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_fwd_add(torch::Tensor x, torch::Tensor res,
+                                           torch::Tensor w, double eps);
+std::vector<torch::Tensor> rmsnorm_bwd_add(torch::Tensor x, torch::Tensor w,
+                                           torch::Tensor invrms, torch::Tensor dy,
+                                           torch::Tensor dsum);
 std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor x, torch::Tensor w,
                                        torch::Tensor invrms, torch::Tensor dy);
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
@@ -69,6 +74,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (gfx950)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (gfx950)");
+  m.def("rmsnorm_fwd_add", &rmsnorm_fwd_add, "fused residual-add + RMSNorm fwd (gfx950)");
+  m.def("rmsnorm_bwd_add", &rmsnorm_bwd_add, "RMSNorm bwd with +dsum fused into dx (gfx950)");
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (gfx950)");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (gfx950)");
   m.def("rope_fwd", &rope_fwd, "RoPE apply fwd/inverse (gfx950)");

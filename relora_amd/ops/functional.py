@@ -64,6 +64,46 @@ def rmsnorm(x, weight, eps=1e-6):
     return rmsnorm_torch(x, weight, eps)
 
 
+class _HipAddRMSNorm(torch.autograd.Function):
+    """Fused `sum = x + residual; normed = rmsnorm(sum)` (K16).  Backward
+    returns the SAME gradient tensor for both inputs (d sum/dx = d sum/dres
+    = 1) with the residual fork's `+dsum` folded into the dx epilogue — the
+    unfused form costs one [M,H] add kernel in each direction."""
+
+    @staticmethod
+    def forward(ctx, x, residual, weight, eps):
+        shape = x.shape
+        x2d = x.contiguous().view(-1, shape[-1])
+        r2d = residual.contiguous().view(-1, shape[-1])
+        y, s, invrms = hip.ext().rmsnorm_fwd_add(x2d, r2d, weight, eps)
+        ctx.save_for_backward(s, weight, invrms)
+        ctx.shape = shape
+        return y.view(shape), s.view(shape)
+
+    @staticmethod
+    def backward(ctx, dy, dsum):
+        s, weight, invrms = ctx.saved_tensors
+        H = s.shape[-1]
+        dy2d = dy.contiguous().view(-1, H)
+        if dsum is None:
+            dx, dw = hip.ext().rmsnorm_bwd(s, weight, invrms, dy2d)
+        else:
+            dx, dw = hip.ext().rmsnorm_bwd_add(s, weight, invrms, dy2d,
+                                               dsum.contiguous().view(-1, H))
+        dx = dx.view(ctx.shape)
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rmsnorm(x, residual, weight, eps=1e-6):
+    """(normed, sum) where sum = x + residual, normed = rmsnorm(sum).
+    Same bf16 rounding as the unfused composition."""
+    if hip.use_hip(x, "rmsnorm") and x.dtype == torch.bfloat16 \
+            and residual.dtype == torch.bfloat16:
+        return _HipAddRMSNorm.apply(x, residual, weight, eps)
+    s = x + residual
+    return rmsnorm(s, weight, eps), s
+
+
 # ---------------------------------------------------------------------------
 # LayerNorm (K6) — pythia path
 # ---------------------------------------------------------------------------

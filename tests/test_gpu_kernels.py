@@ -567,3 +567,33 @@ def test_pythia_train_smoke_gpu():
         losses.append(float(loss))
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0] - 0.5, losses
+
+
+def test_add_rmsnorm_fused_matches_composition():
+    """K16: fused residual-add + RMSNorm vs the unfused composition,
+    forward and backward (both inputs get the fork gradient)."""
+    from relora_amd.ops.functional import _HipAddRMSNorm
+
+    torch.manual_seed(0)
+    M, H = 128, 2048
+    x = torch.randn(M, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    r = torch.randn(M, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y, s = _HipAddRMSNorm.apply(x, r, w, 1e-6)
+
+    xf = x.detach().float().requires_grad_(True)
+    rf = r.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    sf = xf + rf
+    var = sf.pow(2).mean(-1, keepdim=True)
+    yf = wf * (sf * torch.rsqrt(var + 1e-6))
+    assert_close_bf16(y, yf.detach(), what="add_rmsnorm y")
+    assert_close_bf16(s, sf.detach(), what="add_rmsnorm sum")
+
+    dy = torch.randn_like(y)
+    ds = torch.randn_like(s)
+    (y * dy + s * ds).sum().backward()
+    (yf * dy.float() + sf * ds.float()).sum().backward()
+    assert_close_bf16(x.grad, xf.grad, atol=3e-2, rtol=3e-2, what="add_rmsnorm dx")
+    assert_close_bf16(r.grad, rf.grad, atol=3e-2, rtol=3e-2, what="add_rmsnorm dres")
+    assert_close_bf16(w.grad.float(), wf.grad, atol=2e-1, rtol=5e-2, what="add_rmsnorm dw")
