@@ -150,22 +150,32 @@ class LlamaDecoderLayer(nn.Module):
         self.input_layernorm = LlamaRMSNorm(config.hidden_size, eps=config.rms_norm_eps)
         self.post_attention_layernorm = LlamaRMSNorm(config.hidden_size, eps=config.rms_norm_eps)
 
-    def forward(self, hidden_states, position_ids=None, past_key_value=None, use_cache=False):
-        residual = hidden_states
-        hidden_states = self.input_layernorm(hidden_states)
-        hidden_states, present = self.self_attn(
-            hidden_states, position_ids=position_ids,
+    def forward(self, hidden_states, position_ids=None, past_key_value=None,
+                use_cache=False, pending_residual=None, defer_add=False):
+        """Residual adds fuse into the norms (K16, same bf16 rounding as the
+        unfused composition).  With `defer_add`, the MLP add is handed to
+        the NEXT layer's input norm (or the final norm) as
+        `pending_residual`, so every per-layer [M,H] add kernel disappears;
+        returns (out, pending, present) where `pending is None` iff the
+        output is already summed."""
+        if pending_residual is not None:
+            normed, residual = ops.add_rmsnorm(
+                hidden_states, pending_residual, self.input_layernorm.weight,
+                self.input_layernorm.variance_epsilon)
+        else:
+            residual = hidden_states
+            normed = self.input_layernorm(hidden_states)
+        attn_out, present = self.self_attn(
+            normed, position_ids=position_ids,
             past_key_value=past_key_value, use_cache=use_cache,
         )
-        # fused residual add + norm (K16): same bf16 rounding as the
-        # unfused `residual + h` followed by rmsnorm; one kernel fewer in
-        # each direction on GPU
-        normed, residual = ops.add_rmsnorm(
-            hidden_states, residual, self.post_attention_layernorm.weight,
+        normed2, residual2 = ops.add_rmsnorm(
+            attn_out, residual, self.post_attention_layernorm.weight,
             self.post_attention_layernorm.variance_epsilon)
-        hidden_states = self.mlp(normed)
-        hidden_states = residual + hidden_states
-        return hidden_states, present
+        mlp_out = self.mlp(normed2)
+        if defer_add:
+            return mlp_out, residual2, present
+        return mlp_out + residual2, None, present
 
 
 class LlamaPreTrainedModel(PreTrainedModel):
@@ -233,24 +243,34 @@ class LlamaModel(LlamaPreTrainedModel):
 
         all_hidden_states = [] if output_hidden_states else None
         next_cache = [] if use_cache else None
+        # cross-layer K16 fusion: each layer's MLP residual add is deferred
+        # into the next norm (disabled when per-layer hidden states must be
+        # materialized for the output)
+        defer = not output_hidden_states
+        pending = None
         for i, layer in enumerate(self.layers):
             if output_hidden_states:
                 all_hidden_states.append(hidden_states)
             past = past_key_values[i] if past_key_values is not None else None
             if self.gradient_checkpointing and self.training:
-                hidden_states, present = torch.utils.checkpoint.checkpoint(
+                hidden_states, pending, present = torch.utils.checkpoint.checkpoint(
                     layer, hidden_states, position_ids, past, False,
-                    use_reentrant=False,
+                    pending, defer, use_reentrant=False,
                 )
             else:
-                hidden_states, present = layer(
+                hidden_states, pending, present = layer(
                     hidden_states, position_ids=position_ids,
                     past_key_value=past, use_cache=use_cache,
+                    pending_residual=pending, defer_add=defer,
                 )
             if use_cache:
                 next_cache.append(present)
 
-        hidden_states = self.norm(hidden_states)
+        if pending is not None:
+            hidden_states, _ = ops.add_rmsnorm(
+                hidden_states, pending, self.norm.weight, self.norm.variance_epsilon)
+        else:
+            hidden_states = self.norm(hidden_states)
         if output_hidden_states:
             all_hidden_states.append(hidden_states)
 
