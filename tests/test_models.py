@@ -382,3 +382,23 @@ def test_cache_compat_unit():
     assert isinstance(rebuilt, DynamicCache)
     assert rebuilt.get_seq_length() == 3
     assert cache_like([], DynamicCache()) is None
+
+
+def test_llama_deferred_residual_matches_materialized():
+    """The cross-layer residual deferral (defer_add) must be numerically
+    identical to the materialized path (output_hidden_states=True disables
+    the deferral), and hidden-states output must still be per-layer sums."""
+    import torch
+
+    from relora_amd.models import build_model_from_config, load_model_config
+
+    torch.manual_seed(0)
+    cfg = load_model_config("configs/llama_9m.json")
+    model = build_model_from_config(cfg).eval()
+    x = torch.randint(0, cfg.vocab_size, (2, 16))
+    with torch.no_grad():
+        out_deferred = model.model(input_ids=x)  # defer path (no hidden states)
+        out_material = model.model(input_ids=x, output_hidden_states=True)
+    assert torch.equal(out_deferred.last_hidden_state,
+                       out_material.last_hidden_state)
+    assert len(out_material.hidden_states) == cfg.num_hidden_layers + 1
