@@ -739,9 +739,9 @@ DEV_INLINE f32x4 load_f32x4_guard(const float* p, long idx, long n) {
   return r;
 }
 
-template <int HD, int NWAVE = 8>  // NWAVE*32 q rows per block; 4-wave blocks
-                                  // lift occupancy to 3 blocks/CU at hd64
-__global__ __launch_bounds__(NWAVE * 64) void attn_bwd_dq_v3_kernel(
+template <int HD, int NWAVE = 8, int MINW = 2>  // NWAVE*32 q rows per block;
+                                  // MINW=4 caps VGPRs at 128 (occ experiment)
+__global__ __launch_bounds__(NWAVE * 64, MINW) void attn_bwd_dq_v3_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -1542,7 +1542,19 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k, torch::Ten
         const char* e = getenv("RELORA_AMD_DQ_4WAVE");
         return e && e[0] == '1';
       }();
-      if (dq4w && HD <= 64)
+      static const bool dqocc4 = [] {
+        const char* e = getenv("RELORA_AMD_DQ_OCC4");
+        return e && e[0] == '1';
+      }();
+      if (dqocc4 && HD <= 64)
+        hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD, 8, 4>), dim3((S + 255) / 256, B * nh),
+                           block, smem_dq3, stream,
+                           (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                           (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),
+                           (__hip_bfloat16*)dq.data_ptr(), S, hd, nh, bst, hst, ld,
+                           (float)scale);
+      else if (dq4w && HD <= 64)
         hipLaunchKernelGGL((attn_bwd_dq_v3_kernel<HD, 4>), dim3((S + 127) / 128, B * nh),
                            dim3(256), smem_dq3, stream,
                            (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
