@@ -68,7 +68,13 @@ class _HipAddRMSNorm(torch.autograd.Function):
     """Fused `sum = x + residual; normed = rmsnorm(sum)` (K16).  Backward
     returns the SAME gradient tensor for both inputs (d sum/dx = d sum/dres
     = 1) with the residual fork's `+dsum` folded into the dx epilogue — the
-    unfused form costs one [M,H] add kernel in each direction."""
+    unfused form costs one [M,H] add kernel in each direction.
+
+    Note: because the two returned gradients alias one tensor, calling this
+    on two LEAF tensors makes their `.grad`s aliases as well (fine for
+    values; an in-place edit of one would show in the other).  The model
+    only ever feeds activations, whose gradients are consumed, not stored.
+    """
 
     @staticmethod
     def forward(ctx, x, residual, weight, eps):
